@@ -1,0 +1,105 @@
+#!/usr/bin/env python3
+"""Generate the committed golden fixtures under tests/golden/.
+
+Run IN THE BUILD CONTAINER (not on the GPU box): the scrypt vectors come from
+Python hashlib (OpenSSL), the independent pin for our oracle and HIP kernels.
+The label vectors additionally depend on the oracle's blake3 commitment
+(pinned separately by the official BLAKE3 vectors in test_kats.py), so they
+pin "scrypt-of-commitment" end to end against OpenSSL.
+
+Reference semantics being pinned: label_i =
+scrypt(P=commitment||LE64(i), S="", N, r=1, p=1, dkLen=32) — the post-rs
+CpuInitializer computation reached via activation/post.go:295 (RESTATED
+layout; see oracle/oracle.h).
+"""
+import hashlib
+import json
+import os
+import struct
+import sys
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, os.path.join(HERE, "..", ".."))
+sys.path.insert(0, os.path.join(HERE, "..", "..", "oracle"))
+
+NODE_ID = bytes([0xA5]) * 32
+ATX_ID = bytes([0x5A]) * 32
+
+
+def main():
+    from oracle import Oracle
+    o = Oracle()
+
+    out = {}
+
+    # RFC 7914 §12 scrypt KATs (values verified against the RFC text).
+    out["scrypt_rfc7914"] = [
+        {"P": "", "S": "", "N": 16, "r": 1, "p": 1, "dkLen": 64,
+         "out": "77d6576238657b203b19ca42c18a0497f16b4844e3074ae8dfdffa3fe"
+                "de21442fcd0069ded0948f8326a753a0fc81f17e8d3e0fb2e0d3628cf3"
+                "5e20c38d1890"
+                "6"},
+        {"P": "password", "S": "NaCl", "N": 1024, "r": 8, "p": 16,
+         "dkLen": 64,
+         "out": "fdbabe1c9d3472007856e7190d01e9fe7c6ad7cbc8237830e77376634b"
+                "3731622eaf30d92e22a3886ff109279d9830dac727afb94a83ee6d8360"
+                "cbdfa2cc0640"},
+        {"P": "pleaseletmein", "S": "SodiumChloride", "N": 16384, "r": 8,
+         "p": 1, "dkLen": 64,
+         "out": "7023bdcb3afd7348461c06cd81fd38ebfda8fbba904f8e3ea9b543f654"
+                "5da1f2d5432955613f0fcf62d49705242a9af9e61e85dc0d651e40dfcf"
+                "017b45575887"},
+    ]
+
+    # Randomized (fixed-seed) scrypt vectors at post parameters, from OpenSSL.
+    post_vectors = []
+    for i in range(8):
+        pw = hashlib.sha256(b"post-vector-%d" % i).digest() + \
+            hashlib.sha256(b"post-vector-b-%d" % i).digest()[:8]  # 40 B
+        for n in (2, 32, 8192):
+            ref = hashlib.scrypt(pw, salt=b"", n=n, r=1, p=1, dklen=32,
+                                 maxmem=2**26)
+            post_vectors.append({"P_hex": pw.hex(), "N": n,
+                                 "out": ref.hex()})
+    out["scrypt_post_params_openssl"] = post_vectors
+
+    # Official BLAKE3 vectors (from the BLAKE3 reference test_vectors.json;
+    # the two inputs used here are the empty input and the single byte 0x00 —
+    # the input pattern of the official file for length 1).
+    out["blake3_official"] = [
+        {"input_hex": "",
+         "out": "af1349b9f5f9a1a6a0404dea36dcc9499bcb25c9adc112b7cc9a93cae4"
+                "1f3262"},
+        {"input_hex": "00",
+         "out": "2d3adedff11b61f14c886e35afa036736dcd87a74d27b5c1510225d0f5"
+                "92e213"},
+    ]
+
+    # FIPS-197 Appendix C.1
+    out["aes128_fips197"] = [
+        {"key": "000102030405060708090a0b0c0d0e0f",
+         "pt": "00112233445566778899aabbccddeeff",
+         "ct": "69c4e0d86a7b0430d8cdb78070b4c55a"},
+    ]
+
+    # End-to-end label vectors: commitment from oracle blake3, scrypt from
+    # OpenSSL.  These pin the oracle AND the HIP kernel's full label path.
+    commitment = o.commitment(NODE_ID, ATX_ID)
+    label_vectors = {"node_id": NODE_ID.hex(), "atx_id": ATX_ID.hex(),
+                     "commitment": commitment.hex(), "labels": []}
+    for n in (2, 128, 8192):
+        for idx in (0, 1, 2, 63, 64, 1000, 2**32 + 17):
+            pw = commitment + struct.pack("<Q", idx)
+            full = hashlib.scrypt(pw, salt=b"", n=n, r=1, p=1, dklen=32,
+                                  maxmem=2**26)
+            label_vectors["labels"].append(
+                {"N": n, "index": idx, "full": full.hex()})
+    out["labels_openssl"] = label_vectors
+
+    with open(os.path.join(HERE, "golden.json"), "w") as f:
+        json.dump(out, f, indent=1)
+    print("wrote", os.path.join(HERE, "golden.json"))
+
+
+if __name__ == "__main__":
+    main()
