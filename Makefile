@@ -1,0 +1,27 @@
+# Build the MI355X POST engine (libpost_hip.so, gfx950) and the CPU oracle.
+# hipcc cross-compiles gfx950 without a GPU; the in-tree .so travels to the
+# GPU box with the gpurun snapshot.
+HIPCC   ?= hipcc
+ARCH    ?= gfx950
+HIPFLAGS ?= --offload-arch=$(ARCH) -O3 -std=c++17 -fPIC -Wall
+
+ENGINE  := go-spacemesh_amd/libpost_hip.so
+CSRC    := go-spacemesh_amd/csrc
+
+all: $(ENGINE) oracle
+
+$(ENGINE): $(CSRC)/kernels.hip $(CSRC)/engine.cpp $(CSRC)/crypto_host.cpp \
+           $(CSRC)/post_common.h $(CSRC)/kernel_args.h $(CSRC)/crypto_host.h \
+           include/spacemesh_post.h
+	$(HIPCC) $(HIPFLAGS) -shared \
+	    $(CSRC)/kernels.hip $(CSRC)/engine.cpp $(CSRC)/crypto_host.cpp \
+	    -o $@
+
+oracle:
+	$(MAKE) -C oracle
+
+clean:
+	rm -f $(ENGINE)
+	$(MAKE) -C oracle clean
+
+.PHONY: all oracle clean

@@ -1,0 +1,505 @@
+/* crypto_host.cpp — engine host crypto (see crypto_host.h).  Independent
+ * implementation #2; cross-checked in tests against the oracle and the
+ * committed OpenSSL/FIPS/BLAKE3 golden vectors. */
+#include "crypto_host.h"
+#include "post_common.h"
+
+#include <atomic>
+#include <cstring>
+#include <thread>
+#include <vector>
+
+namespace poste {
+
+/* ------------------------------ BLAKE3 ------------------------------ */
+/* Single-chunk (<=1024 B) implementation — every input on this path is
+ * <=64 bytes (commitment) or <=49 bytes (k2pow/cipher keys). */
+namespace b3 {
+
+static constexpr uint32_t IV[8] = {0x6a09e667u, 0xbb67ae85u, 0x3c6ef372u,
+                                   0xa54ff53au, 0x510e527fu, 0x9b05688cu,
+                                   0x1f83d9abu, 0x5be0cd19u};
+static constexpr uint32_t F_CHUNK_START = 1, F_CHUNK_END = 2, F_ROOT = 8;
+
+static inline uint32_t ror(uint32_t x, int n) {
+  return (x >> n) | (x << (32 - n));
+}
+
+static void round_fn(uint32_t v[16], const uint32_t m[16]) {
+  static constexpr int SCHED[8][4] = {{0, 4, 8, 12},  {1, 5, 9, 13},
+                                      {2, 6, 10, 14}, {3, 7, 11, 15},
+                                      {0, 5, 10, 15}, {1, 6, 11, 12},
+                                      {2, 7, 8, 13},  {3, 4, 9, 14}};
+  for (int i = 0; i < 8; i++) {
+    const int a = SCHED[i][0], b = SCHED[i][1], c = SCHED[i][2],
+              d = SCHED[i][3];
+    v[a] += v[b] + m[2 * i];
+    v[d] = ror(v[d] ^ v[a], 16);
+    v[c] += v[d];
+    v[b] = ror(v[b] ^ v[c], 12);
+    v[a] += v[b] + m[2 * i + 1];
+    v[d] = ror(v[d] ^ v[a], 8);
+    v[c] += v[d];
+    v[b] = ror(v[b] ^ v[c], 7);
+  }
+}
+
+static void compress(const uint32_t h[8], const uint32_t m_in[16], uint64_t t,
+                     uint32_t blen, uint32_t flags, uint32_t out[16]) {
+  static constexpr uint8_t P[16] = {2, 6,  3,  10, 7, 0,  4,  13,
+                                    1, 11, 12, 5,  9, 14, 15, 8};
+  uint32_t v[16] = {h[0],  h[1],  h[2],        h[3],
+                    h[4],  h[5],  h[6],        h[7],
+                    IV[0], IV[1], IV[2],       IV[3],
+                    (uint32_t)t, (uint32_t)(t >> 32), blen, flags};
+  uint32_t m[16];
+  std::memcpy(m, m_in, 64);
+  for (int r = 0;; r++) {
+    round_fn(v, m);
+    if (r == 6) break;
+    uint32_t nm[16];
+    for (int i = 0; i < 16; i++) nm[i] = m[P[i]];
+    std::memcpy(m, nm, 64);
+  }
+  for (int i = 0; i < 8; i++) {
+    out[i] = v[i] ^ v[i + 8];
+    out[i + 8] = v[i + 8] ^ h[i];
+  }
+}
+
+} // namespace b3
+
+void blake3(const uint8_t *msg, size_t len, uint8_t *out, size_t outlen) {
+  /* chain full 64-byte blocks, keep the last for the root compressions */
+  uint32_t h[8];
+  std::memcpy(h, b3::IV, 32);
+  size_t nblocks = len == 0 ? 1 : (len + 63) / 64;
+  uint32_t tmp[16];
+  for (size_t b = 0; b + 1 < nblocks; b++) {
+    uint32_t m[16];
+    std::memcpy(m, msg + 64 * b, 64); /* x86/gfx: little-endian host */
+    b3::compress(h, m, 0, 64, b == 0 ? b3::F_CHUNK_START : 0, tmp);
+    std::memcpy(h, tmp, 32);
+  }
+  uint8_t last[64] = {0};
+  uint32_t llen = (uint32_t)(len - (nblocks - 1) * 64);
+  std::memcpy(last, msg + (nblocks - 1) * 64, llen);
+  uint32_t m[16];
+  std::memcpy(m, last, 64);
+  uint32_t flags = (nblocks == 1 ? b3::F_CHUNK_START : 0) | b3::F_CHUNK_END |
+                   b3::F_ROOT;
+  uint64_t t = 0;
+  size_t off = 0;
+  while (off < outlen) {
+    b3::compress(h, m, t++, llen, flags, tmp);
+    size_t take = outlen - off < 64 ? outlen - off : 64;
+    std::memcpy(out + off, tmp, take);
+    off += take;
+  }
+}
+
+/* ------------------------------ SHA-256 ------------------------------ */
+namespace sh {
+static constexpr uint32_t K[64] = {
+    0x428a2f98, 0x71374491, 0xb5c0fbcf, 0xe9b5dba5, 0x3956c25b, 0x59f111f1,
+    0x923f82a4, 0xab1c5ed5, 0xd807aa98, 0x12835b01, 0x243185be, 0x550c7dc3,
+    0x72be5d74, 0x80deb1fe, 0x9bdc06a7, 0xc19bf174, 0xe49b69c1, 0xefbe4786,
+    0x0fc19dc6, 0x240ca1cc, 0x2de92c6f, 0x4a7484aa, 0x5cb0a9dc, 0x76f988da,
+    0x983e5152, 0xa831c66d, 0xb00327c8, 0xbf597fc7, 0xc6e00bf3, 0xd5a79147,
+    0x06ca6351, 0x14292967, 0x27b70a85, 0x2e1b2138, 0x4d2c6dfc, 0x53380d13,
+    0x650a7354, 0x766a0abb, 0x81c2c92e, 0x92722c85, 0xa2bfe8a1, 0xa81a664b,
+    0xc24b8b70, 0xc76c51a3, 0xd192e819, 0xd6990624, 0xf40e3585, 0x106aa070,
+    0x19a4c116, 0x1e376c08, 0x2748774c, 0x34b0bcb5, 0x391c0cb3, 0x4ed8aa4a,
+    0x5b9cca4f, 0x682e6ff3, 0x748f82ee, 0x78a5636f, 0x84c87814, 0x8cc70208,
+    0x90befffa, 0xa4506ceb, 0xbef9a3f7, 0xc67178f2};
+static inline uint32_t ror(uint32_t x, int n) {
+  return (x >> n) | (x << (32 - n));
+}
+static void compress(uint32_t s[8], const uint8_t blk[64]) {
+  uint32_t w[16], a = s[0], b = s[1], c = s[2], d = s[3], e = s[4], f = s[5],
+               g = s[6], h = s[7];
+  for (int i = 0; i < 16; i++)
+    w[i] = ((uint32_t)blk[4 * i] << 24) | ((uint32_t)blk[4 * i + 1] << 16) |
+           ((uint32_t)blk[4 * i + 2] << 8) | blk[4 * i + 3];
+  for (int t = 0; t < 64; t++) {
+    uint32_t wt;
+    if (t < 16) {
+      wt = w[t];
+    } else {
+      uint32_t a15 = w[(t - 15) & 15], a2 = w[(t - 2) & 15];
+      wt = w[t & 15] += (ror(a15, 7) ^ ror(a15, 18) ^ (a15 >> 3)) +
+                        w[(t - 7) & 15] +
+                        (ror(a2, 17) ^ ror(a2, 19) ^ (a2 >> 10));
+    }
+    uint32_t t1 = h + (ror(e, 6) ^ ror(e, 11) ^ ror(e, 25)) +
+                  ((e & f) ^ (~e & g)) + K[t] + wt;
+    uint32_t t2 = (ror(a, 2) ^ ror(a, 13) ^ ror(a, 22)) +
+                  ((a & b) ^ (a & c) ^ (b & c));
+    h = g; g = f; f = e; e = d + t1;
+    d = c; c = b; b = a; a = t1 + t2;
+  }
+  s[0] += a; s[1] += b; s[2] += c; s[3] += d;
+  s[4] += e; s[5] += f; s[6] += g; s[7] += h;
+}
+} // namespace sh
+
+void sha256(const uint8_t *msg, size_t len, uint8_t out[32]) {
+  uint32_t s[8] = {0x6a09e667, 0xbb67ae85, 0x3c6ef372, 0xa54ff53a,
+                   0x510e527f, 0x9b05688c, 0x1f83d9ab, 0x5be0cd19};
+  size_t i = 0;
+  for (; i + 64 <= len; i += 64) sh::compress(s, msg + i);
+  uint8_t tail[128] = {0};
+  size_t rem = len - i;
+  std::memcpy(tail, msg + i, rem);
+  tail[rem] = 0x80;
+  size_t tlen = rem + 9 <= 64 ? 64 : 128;
+  uint64_t bits = (uint64_t)len * 8;
+  for (int b = 0; b < 8; b++)
+    tail[tlen - 8 + b] = (uint8_t)(bits >> (56 - 8 * b));
+  sh::compress(s, tail);
+  if (tlen == 128) sh::compress(s, tail + 64);
+  for (int k = 0; k < 8; k++) {
+    out[4 * k] = (uint8_t)(s[k] >> 24);
+    out[4 * k + 1] = (uint8_t)(s[k] >> 16);
+    out[4 * k + 2] = (uint8_t)(s[k] >> 8);
+    out[4 * k + 3] = (uint8_t)s[k];
+  }
+}
+
+/* HMAC-SHA256 with key <= 64 bytes (all keys on this path are 40 or 128
+ * bytes; 128-byte salts are message-side).  General length via pre-hash. */
+static void hmac256(const uint8_t *key, size_t klen, const uint8_t *m1,
+                    size_t l1, const uint8_t *m2, size_t l2,
+                    uint8_t out[32]) {
+  uint8_t k[64] = {0};
+  if (klen > 64)
+    sha256(key, klen, k);
+  else
+    std::memcpy(k, key, klen);
+  std::vector<uint8_t> buf(64 + l1 + l2);
+  for (int i = 0; i < 64; i++) buf[i] = k[i] ^ 0x36;
+  std::memcpy(buf.data() + 64, m1, l1);
+  if (l2) std::memcpy(buf.data() + 64 + l1, m2, l2);
+  uint8_t inner[32];
+  sha256(buf.data(), buf.size(), inner);
+  uint8_t obuf[96];
+  for (int i = 0; i < 64; i++) obuf[i] = k[i] ^ 0x5c;
+  std::memcpy(obuf + 64, inner, 32);
+  sha256(obuf, 96, out);
+}
+
+int scrypt_r1p1(const uint8_t *pass, size_t passlen, uint32_t n,
+                uint8_t out[32]) {
+  if (n < 2 || (n & (n - 1))) return -1;
+  /* B = PBKDF2(P, "", 1, 128): 4 blocks of HMAC(P, "" || INT(i)) */
+  uint8_t B[128];
+  for (uint32_t i = 1; i <= 4; i++) {
+    uint8_t ib[4] = {(uint8_t)(i >> 24), (uint8_t)(i >> 16), (uint8_t)(i >> 8),
+                     (uint8_t)i};
+    hmac256(pass, passlen, ib, 4, nullptr, 0, B + 32 * (i - 1));
+  }
+  uint32_t X[32];
+  for (int k = 0; k < 32; k++)
+    X[k] = (uint32_t)B[4 * k] | ((uint32_t)B[4 * k + 1] << 8) |
+           ((uint32_t)B[4 * k + 2] << 16) | ((uint32_t)B[4 * k + 3] << 24);
+  std::vector<uint32_t> V((size_t)n * 32);
+  auto salsa8 = [](uint32_t *b) {
+    uint32_t x[16];
+    std::memcpy(x, b, 64);
+    auto R = [](uint32_t v, int s) { return (v << s) | (v >> (32 - s)); };
+    for (int i = 0; i < 4; i++) {
+      x[4] ^= R(x[0] + x[12], 7);   x[8] ^= R(x[4] + x[0], 9);
+      x[12] ^= R(x[8] + x[4], 13);  x[0] ^= R(x[12] + x[8], 18);
+      x[9] ^= R(x[5] + x[1], 7);    x[13] ^= R(x[9] + x[5], 9);
+      x[1] ^= R(x[13] + x[9], 13);  x[5] ^= R(x[1] + x[13], 18);
+      x[14] ^= R(x[10] + x[6], 7);  x[2] ^= R(x[14] + x[10], 9);
+      x[6] ^= R(x[2] + x[14], 13);  x[10] ^= R(x[6] + x[2], 18);
+      x[3] ^= R(x[15] + x[11], 7);  x[7] ^= R(x[3] + x[15], 9);
+      x[11] ^= R(x[7] + x[3], 13);  x[15] ^= R(x[11] + x[7], 18);
+      x[1] ^= R(x[0] + x[3], 7);    x[2] ^= R(x[1] + x[0], 9);
+      x[3] ^= R(x[2] + x[1], 13);   x[0] ^= R(x[3] + x[2], 18);
+      x[6] ^= R(x[5] + x[4], 7);    x[7] ^= R(x[6] + x[5], 9);
+      x[4] ^= R(x[7] + x[6], 13);   x[5] ^= R(x[4] + x[7], 18);
+      x[11] ^= R(x[10] + x[9], 7);  x[8] ^= R(x[11] + x[10], 9);
+      x[9] ^= R(x[8] + x[11], 13);  x[10] ^= R(x[9] + x[8], 18);
+      x[12] ^= R(x[15] + x[14], 7); x[13] ^= R(x[12] + x[15], 9);
+      x[14] ^= R(x[13] + x[12], 13); x[15] ^= R(x[14] + x[13], 18);
+    }
+    for (int i = 0; i < 16; i++) b[i] += x[i];
+  };
+  auto blockmix = [&](uint32_t *Xb) {
+    uint32_t T[16];
+    for (int k = 0; k < 16; k++) T[k] = Xb[k] ^ Xb[16 + k];
+    salsa8(T);
+    uint32_t Y0[16];
+    std::memcpy(Y0, T, 64);
+    for (int k = 0; k < 16; k++) T[k] ^= Xb[16 + k];
+    salsa8(T);
+    std::memcpy(Xb, Y0, 64);
+    std::memcpy(Xb + 16, T, 64);
+  };
+  for (uint32_t i = 0; i < n; i++) {
+    std::memcpy(&V[(size_t)i * 32], X, 128);
+    blockmix(X);
+  }
+  for (uint32_t i = 0; i < n; i++) {
+    uint64_t j = ((uint64_t)X[16] | ((uint64_t)X[17] << 32)) % n;
+    const uint32_t *Vj = &V[(size_t)j * 32];
+    for (int k = 0; k < 32; k++) X[k] ^= Vj[k];
+    blockmix(X);
+  }
+  for (int k = 0; k < 32; k++) {
+    B[4 * k] = (uint8_t)X[k];
+    B[4 * k + 1] = (uint8_t)(X[k] >> 8);
+    B[4 * k + 2] = (uint8_t)(X[k] >> 16);
+    B[4 * k + 3] = (uint8_t)(X[k] >> 24);
+  }
+  uint8_t ib[4] = {0, 0, 0, 1};
+  hmac256(pass, passlen, B, 128, ib, 4, out);
+  return 0;
+}
+
+void commitment(const uint8_t node_id[32], const uint8_t atx_id[32],
+                uint8_t out[32]) {
+  uint8_t buf[64];
+  std::memcpy(buf, node_id, 32);
+  std::memcpy(buf + 32, atx_id, 32);
+  blake3(buf, 64, out, 32);
+}
+
+int host_label(const uint8_t commitment32[32], uint64_t index, uint32_t n,
+               uint8_t out[32]) {
+  uint8_t pass[40];
+  std::memcpy(pass, commitment32, 32);
+  for (int i = 0; i < 8; i++) pass[32 + i] = (uint8_t)(index >> (8 * i));
+  return scrypt_r1p1(pass, 40, n, out);
+}
+
+/* ------------------------------ AES-128 ------------------------------ */
+namespace aes {
+static uint8_t SBOX[256];
+static bool ready = false;
+static inline uint8_t xtime(uint8_t a) {
+  return (uint8_t)((a << 1) ^ ((a & 0x80) ? 0x1b : 0));
+}
+static uint8_t mul(uint8_t a, uint8_t b) {
+  uint8_t r = 0;
+  while (b) {
+    if (b & 1) r ^= a;
+    a = xtime(a);
+    b >>= 1;
+  }
+  return r;
+}
+static void init_sbox() {
+  if (ready) return;
+  /* inverse via a^{254} (Fermat in GF(2^8)) then affine */
+  for (int i = 0; i < 256; i++) {
+    uint8_t a = (uint8_t)i, inv = 0;
+    if (a) {
+      uint8_t p = a, acc = 1;
+      /* 254 = 0b11111110 */
+      for (int bit = 7; bit >= 0; bit--) {
+        acc = mul(acc, acc);
+        if ((254 >> bit) & 1) acc = mul(acc, p);
+      }
+      inv = acc;
+    }
+    uint8_t b = inv;
+    uint8_t s = (uint8_t)(b ^ ((b << 1) | (b >> 7)) ^ ((b << 2) | (b >> 6)) ^
+                          ((b << 3) | (b >> 5)) ^ ((b << 4) | (b >> 4)) ^
+                          0x63);
+    SBOX[i] = s;
+  }
+  ready = true;
+}
+} // namespace aes
+
+void aes128_tables(uint32_t te[1024], uint8_t sbox[256]) {
+  aes::init_sbox();
+  std::memcpy(sbox, aes::SBOX, 256);
+  for (int x = 0; x < 256; x++) {
+    uint8_t s = aes::SBOX[x];
+    uint8_t s2 = aes::xtime(s), s3 = (uint8_t)(s2 ^ s);
+    uint32_t t0 = ((uint32_t)s2 << 24) | ((uint32_t)s << 16) |
+                  ((uint32_t)s << 8) | s3;
+    te[x] = t0;
+    te[256 + x] = (t0 >> 8) | (t0 << 24);
+    te[512 + x] = (t0 >> 16) | (t0 << 16);
+    te[768 + x] = (t0 >> 24) | (t0 << 8);
+  }
+}
+
+void aes128_expand(const uint8_t key[16], uint32_t rk_be[44]) {
+  aes::init_sbox();
+  for (int i = 0; i < 4; i++)
+    rk_be[i] = ((uint32_t)key[4 * i] << 24) | ((uint32_t)key[4 * i + 1] << 16) |
+               ((uint32_t)key[4 * i + 2] << 8) | key[4 * i + 3];
+  uint8_t rcon = 1;
+  for (int i = 4; i < 44; i++) {
+    uint32_t t = rk_be[i - 1];
+    if (i % 4 == 0) {
+      t = (t << 8) | (t >> 24); /* RotWord */
+      t = ((uint32_t)aes::SBOX[(t >> 24) & 0xff] << 24) |
+          ((uint32_t)aes::SBOX[(t >> 16) & 0xff] << 16) |
+          ((uint32_t)aes::SBOX[(t >> 8) & 0xff] << 8) |
+          aes::SBOX[t & 0xff];
+      t ^= (uint32_t)rcon << 24;
+      rcon = aes::xtime(rcon);
+    }
+    rk_be[i] = rk_be[i - 4] ^ t;
+  }
+}
+
+void aes128_enc_block(const uint32_t rk[44], const uint8_t in[16],
+                      uint8_t out[16]) {
+  static uint32_t TE[1024];
+  static uint8_t SB[256];
+  static bool tbl = false;
+  if (!tbl) {
+    aes128_tables(TE, SB);
+    tbl = true;
+  }
+  uint32_t w[4];
+  for (int j = 0; j < 4; j++)
+    w[j] = (((uint32_t)in[4 * j] << 24) | ((uint32_t)in[4 * j + 1] << 16) |
+            ((uint32_t)in[4 * j + 2] << 8) | in[4 * j + 3]) ^ rk[j];
+  for (int r = 1; r < 10; r++) {
+    uint32_t n[4];
+    for (int j = 0; j < 4; j++)
+      n[j] = TE[w[j] >> 24] ^ TE[256 + ((w[(j + 1) & 3] >> 16) & 0xff)] ^
+             TE[512 + ((w[(j + 2) & 3] >> 8) & 0xff)] ^
+             TE[768 + (w[(j + 3) & 3] & 0xff)] ^ rk[4 * r + j];
+    std::memcpy(w, n, 16);
+  }
+  uint32_t f[4];
+  for (int j = 0; j < 4; j++)
+    f[j] = (((uint32_t)SB[w[j] >> 24] << 24) |
+            ((uint32_t)SB[(w[(j + 1) & 3] >> 16) & 0xff] << 16) |
+            ((uint32_t)SB[(w[(j + 2) & 3] >> 8) & 0xff] << 8) |
+            SB[w[(j + 3) & 3] & 0xff]) ^ rk[40 + j];
+  for (int j = 0; j < 4; j++) {
+    out[4 * j] = (uint8_t)(f[j] >> 24);
+    out[4 * j + 1] = (uint8_t)(f[j] >> 16);
+    out[4 * j + 2] = (uint8_t)(f[j] >> 8);
+    out[4 * j + 3] = (uint8_t)f[j];
+  }
+}
+
+/* ------------------------- protocol derivations ------------------------- */
+void prove_cipher_key(const uint8_t challenge[32], uint32_t cipher,
+                      uint64_t group_pow, uint8_t out[16]) {
+  uint8_t msg[44];
+  std::memcpy(msg, challenge, 32);
+  for (int i = 0; i < 4; i++) msg[32 + i] = (uint8_t)(cipher >> (8 * i));
+  for (int i = 0; i < 8; i++) msg[36 + i] = (uint8_t)(group_pow >> (8 * i));
+  uint8_t h[32];
+  blake3(msg, 44, h, 32);
+  std::memcpy(out, h, 16);
+}
+
+int k2pow_verify_blake3(const uint8_t challenge[32], uint32_t nonce_group,
+                        uint64_t pow, const uint8_t pow_difficulty[32]) {
+  uint8_t msg[49], h[32];
+  std::memcpy(msg, POSTE_K2POW_PREFIX, 5);
+  std::memcpy(msg + 5, challenge, 32);
+  for (int i = 0; i < 4; i++) msg[37 + i] = (uint8_t)(nonce_group >> (8 * i));
+  for (int i = 0; i < 8; i++) msg[41 + i] = (uint8_t)(pow >> (8 * i));
+  blake3(msg, 49, h, 32);
+  return std::memcmp(h, pow_difficulty, 32) < 0 ? 0 : -1;
+}
+
+uint64_t k2pow_search_blake3(const uint8_t challenge[32], uint32_t nonce_group,
+                             const uint8_t pow_difficulty[32],
+                             uint32_t threads) {
+  if (threads == 0) threads = std::thread::hardware_concurrency();
+  if (threads == 0) threads = 1;
+  const uint64_t CHUNK = 16384;
+  for (uint64_t base = 0;; base += CHUNK * threads) {
+    std::atomic<uint64_t> best{UINT64_MAX};
+    std::vector<std::thread> ts;
+    for (uint32_t t = 0; t < threads; t++) {
+      ts.emplace_back([&, t] {
+        uint64_t lo = base + (uint64_t)t * CHUNK;
+        for (uint64_t p = lo; p < lo + CHUNK; p++) {
+          if (k2pow_verify_blake3(challenge, nonce_group, p,
+                                  pow_difficulty) == 0) {
+            uint64_t cur = best.load();
+            while (p < cur && !best.compare_exchange_weak(cur, p)) {
+            }
+            return;
+          }
+        }
+      });
+    }
+    for (auto &th : ts) th.join();
+    if (best.load() != UINT64_MAX) return best.load();
+  }
+}
+
+uint64_t proving_difficulty(uint32_t k1, uint64_t num_labels) {
+  unsigned __int128 d = ((unsigned __int128)k1 << 64) / num_labels;
+  return d > (unsigned __int128)UINT64_MAX ? UINT64_MAX : (uint64_t)d;
+}
+
+void vrf_difficulty(uint64_t num_labels, uint8_t out[32]) {
+  if (num_labels <= 1) {
+    std::memset(out, 0xff, 32);
+    return;
+  }
+  unsigned __int128 rem = 1;
+  for (int limb = 0; limb < 4; limb++) {
+    rem <<= 64;
+    uint64_t q = (uint64_t)(rem / num_labels);
+    rem %= num_labels;
+    for (int b = 0; b < 8; b++)
+      out[limb * 8 + b] = (uint8_t)(q >> (56 - 8 * b));
+  }
+}
+
+uint32_t bits_per_index(uint64_t num_labels) {
+  uint32_t bits = 0;
+  for (uint64_t v = num_labels - 1; v; v >>= 1) bits++;
+  return bits ? bits : 1;
+}
+
+uint32_t pack_indices(const uint64_t *idx, uint32_t k, uint32_t bpi,
+                      uint8_t *out, uint32_t cap) {
+  uint64_t total = (uint64_t)k * bpi;
+  uint32_t bytes = (uint32_t)((total + 7) / 8);
+  if (bytes > cap) return 0;
+  std::memset(out, 0, bytes);
+  uint64_t pos = 0;
+  for (uint32_t i = 0; i < k; i++)
+    for (uint32_t b = 0; b < bpi; b++, pos++)
+      if ((idx[i] >> b) & 1) out[pos >> 3] |= (uint8_t)(1u << (pos & 7));
+  return bytes;
+}
+
+void unpack_indices(const uint8_t *packed, uint32_t k, uint32_t bpi,
+                    uint64_t *idx_out) {
+  uint64_t pos = 0;
+  for (uint32_t i = 0; i < k; i++) {
+    uint64_t v = 0;
+    for (uint32_t b = 0; b < bpi; b++, pos++)
+      if ((packed[pos >> 3] >> (pos & 7)) & 1) v |= (uint64_t)1 << b;
+    idx_out[i] = v;
+  }
+}
+
+void subset_positions(uint32_t k2, uint32_t k3, const uint8_t *seed,
+                      size_t seed_len, uint32_t *positions_out) {
+  std::vector<uint32_t> perm(k2);
+  for (uint32_t i = 0; i < k2; i++) perm[i] = i;
+  std::vector<uint8_t> stream((size_t)k3 * 8);
+  blake3(seed, seed_len, stream.data(), stream.size());
+  for (uint32_t i = 0; i < k3 && i < k2; i++) {
+    uint64_t d = 0;
+    for (int b = 0; b < 8; b++) d |= (uint64_t)stream[i * 8 + b] << (8 * b);
+    uint32_t j = i + (uint32_t)(d % (k2 - i));
+    std::swap(perm[i], perm[j]);
+    positions_out[i] = perm[i];
+  }
+}
+
+} // namespace poste

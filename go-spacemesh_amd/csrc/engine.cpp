@@ -1,0 +1,962 @@
+/* engine.cpp — C-ABI implementation of the MI355X POST engine
+ * (include/spacemesh_post.h).  Host orchestration around the HIP kernels:
+ * sessions, streams, file IO, resume, nonce tracking, proving pipeline,
+ * batched verification.  No CPU compute fallback: every compute entry
+ * requires a HIP device and fails loudly without one (POST_ERR_NO_GPU).
+ *
+ * Reference lifecycle being mirrored (file:line under /root/reference/):
+ *   activation/post.go:261,267-331   init progress/resume/cancel/state
+ *   activation/post.go:299-312       reference-label self-check
+ *   api/grpcserver/post_client.go:69-143  prove protocol results
+ *   activation/post_verifier.go:150-160 + validation.go:182-222  verify
+ */
+#include "../../include/spacemesh_post.h"
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <array>
+#include <atomic>
+#include <cstdio>
+#include <cstring>
+#include <map>
+#include <mutex>
+#include <string>
+#include <vector>
+
+#include "crypto_host.h"
+#include "kernel_args.h"
+#include "post_common.h"
+
+namespace {
+
+thread_local std::string g_last_error;
+
+void set_error(const std::string &e) { g_last_error = e; }
+
+int hip_fail(const char *what, hipError_t err) {
+  set_error(std::string(what) + ": " + hipGetErrorString(err));
+  return err == hipErrorOutOfMemory ? POST_ERR_OOM : POST_ERR;
+}
+
+#define HIP_TRY(expr)                                                          \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) return hip_fail(#expr, _e);                          \
+  } while (0)
+
+int require_gpu(uint32_t provider_id) {
+  int count = 0;
+  hipError_t e = hipGetDeviceCount(&count);
+  if (e != hipSuccess || count == 0) {
+    set_error("no HIP device available (the HIP engine has no CPU fallback)");
+    return POST_ERR_NO_GPU;
+  }
+  if (provider_id >= (uint32_t)count) {
+    set_error("provider id out of range");
+    return POST_ERR_INVALID_ARGS;
+  }
+  e = hipSetDevice((int)provider_id);
+  if (e != hipSuccess) return hip_fail("hipSetDevice", e);
+  return POST_OK;
+}
+
+void load_commitment_words(const uint8_t c[32], uint32_t w[8]) {
+  std::memcpy(w, c, 32); /* little-endian host/device byte stream words */
+}
+
+void difficulty_to_be_words(const uint8_t d[32], uint32_t w[8]) {
+  for (int i = 0; i < 8; i++)
+    w[i] = ((uint32_t)d[4 * i] << 24) | ((uint32_t)d[4 * i + 1] << 16) |
+           ((uint32_t)d[4 * i + 2] << 8) | d[4 * i + 3];
+}
+
+constexpr uint32_t THREADS = 256;
+constexpr uint32_t CAND_CAP = 1 << 16;
+
+struct DeviceTables { /* AES tables resident per device */
+  uint32_t *d_te = nullptr;
+  uint8_t *d_sbox = nullptr;
+};
+std::mutex g_tables_mu;
+std::map<int, DeviceTables> g_tables;
+
+int get_aes_tables(int dev, DeviceTables &out) {
+  std::lock_guard<std::mutex> lk(g_tables_mu);
+  auto it = g_tables.find(dev);
+  if (it != g_tables.end()) {
+    out = it->second;
+    return POST_OK;
+  }
+  static uint32_t te[1024];
+  static uint8_t sbox[256];
+  poste::aes128_tables(te, sbox);
+  DeviceTables t;
+  HIP_TRY(hipMalloc(&t.d_te, sizeof(te)));
+  HIP_TRY(hipMalloc(&t.d_sbox, sizeof(sbox)));
+  HIP_TRY(hipMemcpy(t.d_te, te, sizeof(te), hipMemcpyHostToDevice));
+  HIP_TRY(hipMemcpy(t.d_sbox, sbox, sizeof(sbox), hipMemcpyHostToDevice));
+  g_tables[dev] = t;
+  out = t;
+  return POST_OK;
+}
+
+} // namespace
+
+extern "C" {
+
+const char *post_last_error(void) { return g_last_error.c_str(); }
+
+const char *post_engine_version(void) {
+  return "spacemesh-post-hip 0.1 (gfx950)";
+}
+
+/* ------------------------- providers ------------------------- */
+
+int post_providers(PostProvider *providers, uint32_t cap, uint32_t *count) {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  if (e != hipSuccess) n = 0;
+  *count = (uint32_t)n;
+  for (uint32_t i = 0; i < (uint32_t)n && i < cap; i++) {
+    hipDeviceProp_t prop;
+    if (hipGetDeviceProperties(&prop, (int)i) != hipSuccess) continue;
+    providers[i].id = i;
+    std::snprintf(providers[i].model, sizeof(providers[i].model), "%s",
+                  prop.name);
+    providers[i].device_type = 0;
+    providers[i].memory_bytes = prop.totalGlobalMem;
+    providers[i].performance = 0;
+  }
+  return POST_OK;
+}
+
+/* ------------------------- init session ------------------------- */
+
+struct PostInitSession {
+  PostInitConfig cfg;
+  std::string data_dir;
+  uint64_t range_start = 0, range_end = 0;
+  std::atomic<uint64_t> written{0};
+  std::atomic<bool> cancel{false};
+  uint8_t commitment[32];
+
+  std::mutex nonce_mu;
+  bool nonce_found = false;
+  uint64_t nonce_idx = 0;
+  uint8_t nonce_label[32];
+
+  uint64_t lanes = 0;       /* concurrent labels (scratch slots) */
+  uint64_t batch = 0;       /* labels per launch */
+  uint32_t *d_scratch = nullptr;
+  uint8_t *d_out = nullptr; /* batch output (16 B per label) */
+  uint8_t *d_all = nullptr; /* whole-range output kept on device (small) */
+  bool keep_all = false;
+  PostVrfCandidate *d_cand = nullptr;
+  unsigned int *d_cand_count = nullptr;
+  uint8_t *h_batch = nullptr; /* pinned */
+  hipStream_t stream = nullptr;
+  hipEvent_t ev0 = nullptr, ev1 = nullptr;
+  double last_kernel_ms = 0; /* accumulated over the last post_init_step */
+
+  ~PostInitSession() {
+    if (ev0) (void)hipEventDestroy(ev0);
+    if (ev1) (void)hipEventDestroy(ev1);
+    if (stream) (void)hipStreamDestroy(stream);
+    if (d_scratch) (void)hipFree(d_scratch);
+    if (d_out) (void)hipFree(d_out);
+    if (d_all) (void)hipFree(d_all);
+    if (d_cand) (void)hipFree(d_cand);
+    if (d_cand_count) (void)hipFree(d_cand_count);
+    if (h_batch) (void)hipHostFree(h_batch);
+  }
+};
+
+static uint64_t existing_labels(const std::string &dir, uint64_t per_file,
+                                uint64_t range_start, uint64_t range_total) {
+  /* resume point = number of contiguous complete labels already on disk
+   * (StartSession resume, activation/post.go:267-271) */
+  uint64_t done = 0;
+  for (uint64_t pos = range_start; pos < range_start + range_total;) {
+    uint64_t file_i = pos / per_file;
+    uint64_t in_file = pos % per_file;
+    uint64_t file_cap = std::min(per_file - in_file,
+                                 range_start + range_total - pos);
+    char path[4096];
+    std::snprintf(path, sizeof path, "%s/postdata_%llu.bin", dir.c_str(),
+                  (unsigned long long)file_i);
+    FILE *f = std::fopen(path, "rb");
+    if (!f) break;
+    std::fseek(f, 0, SEEK_END);
+    long sz = std::ftell(f);
+    std::fclose(f);
+    uint64_t labels_here = (uint64_t)(sz < 0 ? 0 : sz) / POST_LABEL_SIZE;
+    if (in_file > labels_here) break;
+    uint64_t usable = std::min(labels_here - in_file, file_cap);
+    done += usable;
+    pos += usable;
+    if (usable < file_cap) break;
+  }
+  return done;
+}
+
+int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
+  if (!cfg || !out) {
+    set_error("null args");
+    return POST_ERR_INVALID_ARGS;
+  }
+  if (cfg->scrypt_n < 2 || (cfg->scrypt_n & (cfg->scrypt_n - 1))) {
+    set_error("scrypt N must be a power of two >= 2");
+    return POST_ERR_INVALID_ARGS;
+  }
+  int rc = require_gpu(cfg->provider_id);
+  if (rc != POST_OK) return rc;
+
+  auto *s = new PostInitSession();
+  s->cfg = *cfg;
+  if (cfg->data_dir) s->data_dir = cfg->data_dir;
+  uint64_t total = (uint64_t)cfg->num_units * cfg->labels_per_unit;
+  s->range_start = cfg->index_start;
+  s->range_end = cfg->index_end ? cfg->index_end : total;
+  if (s->range_end > total || s->range_start >= s->range_end) {
+    delete s;
+    set_error("bad index range");
+    return POST_ERR_INVALID_ARGS;
+  }
+  poste::commitment(cfg->node_id, cfg->commitment_atx_id, s->commitment);
+
+  /* scratch sizing: lanes * 128 * N bytes */
+  size_t free_b = 0, total_b = 0;
+  HIP_TRY(hipMemGetInfo(&free_b, &total_b));
+  uint64_t budget = cfg->scratch_bytes
+                        ? cfg->scratch_bytes
+                        : (uint64_t)((double)free_b * 0.80);
+  uint64_t per_lane = (uint64_t)cfg->scrypt_n * 128;
+  uint64_t lanes = budget / per_lane;
+  uint64_t range = s->range_end - s->range_start;
+  lanes = std::min<uint64_t>(lanes, std::max<uint64_t>(range, THREADS));
+  /* cap lane count: beyond ~8 waves/CU the scratch grows without latency
+   * benefit (256 CU * 2048 threads = 512Ki lanes max) */
+  lanes = std::min<uint64_t>(lanes, 512ull * 1024);
+  lanes = (lanes / THREADS) * THREADS;
+  if (lanes == 0) {
+    delete s;
+    set_error("not enough device memory for one scratch lane block");
+    return POST_ERR_OOM;
+  }
+  s->lanes = lanes;
+  s->batch = std::min<uint64_t>(range, lanes * 4);
+
+  HIP_TRY(hipMalloc(&s->d_scratch, (size_t)lanes * per_lane));
+  uint64_t keep_bytes = range * POST_LABEL_SIZE;
+  s->keep_all = s->data_dir.empty() && keep_bytes <= (8ull << 30);
+  if (s->keep_all) {
+    HIP_TRY(hipMalloc(&s->d_all, (size_t)keep_bytes));
+  }
+  HIP_TRY(hipMalloc(&s->d_out, (size_t)s->batch * POST_LABEL_SIZE));
+  HIP_TRY(hipMalloc(&s->d_cand, sizeof(PostVrfCandidate) * CAND_CAP));
+  HIP_TRY(hipMalloc(&s->d_cand_count, sizeof(unsigned int)));
+  HIP_TRY(hipHostMalloc(&s->h_batch, (size_t)s->batch * POST_LABEL_SIZE));
+  HIP_TRY(hipStreamCreate(&s->stream));
+
+  if (!s->data_dir.empty()) {
+    uint64_t per_file = std::max<uint64_t>(
+        1, s->cfg.max_file_size / POST_LABEL_SIZE);
+    s->written = existing_labels(s->data_dir, per_file, s->range_start, range);
+  }
+  *out = s;
+  return POST_OK;
+}
+
+uint64_t post_init_num_labels_written(const PostInitSession *s) {
+  return s ? s->written.load() : 0;
+}
+
+void post_init_cancel(PostInitSession *s) {
+  if (s) s->cancel = true;
+}
+
+int post_init_nonce(const PostInitSession *s, uint64_t *index,
+                    uint8_t label[32]) {
+  auto *m = const_cast<PostInitSession *>(s);
+  std::lock_guard<std::mutex> lk(m->nonce_mu);
+  if (!s->nonce_found) return POST_ERR;
+  *index = s->nonce_idx;
+  std::memcpy(label, s->nonce_label, 32);
+  return POST_OK;
+}
+
+static int write_batch_files(PostInitSession *s, uint64_t pos, uint64_t count,
+                             const uint8_t *host_labels) {
+  uint64_t per_file =
+      std::max<uint64_t>(1, s->cfg.max_file_size / POST_LABEL_SIZE);
+  uint64_t off = 0;
+  while (off < count) {
+    uint64_t gpos = pos + off;
+    uint64_t file_i = gpos / per_file;
+    uint64_t in_file = gpos % per_file;
+    uint64_t take = std::min(per_file - in_file, count - off);
+    char path[4096];
+    std::snprintf(path, sizeof path, "%s/postdata_%llu.bin",
+                  s->data_dir.c_str(), (unsigned long long)file_i);
+    FILE *f = std::fopen(path, "r+b");
+    if (!f) f = std::fopen(path, "w+b");
+    if (!f) {
+      set_error(std::string("cannot open ") + path);
+      return POST_ERR_IO;
+    }
+    std::fseek(f, (long)(in_file * POST_LABEL_SIZE), SEEK_SET);
+    size_t wr = std::fwrite(host_labels + off * POST_LABEL_SIZE,
+                            POST_LABEL_SIZE, take, f);
+    std::fclose(f);
+    if (wr != take) {
+      set_error("short write");
+      return POST_ERR_IO;
+    }
+    off += take;
+  }
+  return POST_OK;
+}
+
+static void b64(const uint8_t *in, size_t n, std::string &out) {
+  static const char t[] =
+      "ABCDEFGHIJKLMNOPQRSTUVWXYZabcdefghijklmnopqrstuvwxyz0123456789+/";
+  for (size_t i = 0; i < n; i += 3) {
+    uint32_t v = (uint32_t)in[i] << 16;
+    if (i + 1 < n) v |= (uint32_t)in[i + 1] << 8;
+    if (i + 2 < n) v |= in[i + 2];
+    out += t[(v >> 18) & 63];
+    out += t[(v >> 12) & 63];
+    out += i + 1 < n ? t[(v >> 6) & 63] : '=';
+    out += i + 2 < n ? t[v & 63] : '=';
+  }
+}
+
+static int write_metadata(PostInitSession *s) {
+  /* postdata_metadata.json, field set per shared.PostMetadata usage
+   * (activation/post_test.go:305-309; Go json base64 for byte slices) */
+  std::string nid, atx, nv;
+  b64(s->cfg.node_id, 32, nid);
+  b64(s->cfg.commitment_atx_id, 32, atx);
+  char path[4096];
+  std::snprintf(path, sizeof path, "%s/postdata_metadata.json",
+                s->data_dir.c_str());
+  FILE *f = std::fopen(path, "w");
+  if (!f) {
+    set_error("cannot write metadata");
+    return POST_ERR_IO;
+  }
+  std::fprintf(f,
+               "{\n  \"NodeId\": \"%s\",\n  \"CommitmentAtxId\": \"%s\",\n"
+               "  \"LabelsPerUnit\": %llu,\n  \"NumUnits\": %u,\n"
+               "  \"MaxFileSize\": %llu,\n  \"Scrypt\": {\"N\": %u, \"R\": 1,"
+               " \"P\": 1}",
+               nid.c_str(), atx.c_str(),
+               (unsigned long long)s->cfg.labels_per_unit, s->cfg.num_units,
+               (unsigned long long)s->cfg.max_file_size, s->cfg.scrypt_n);
+  std::lock_guard<std::mutex> lk(s->nonce_mu);
+  if (s->nonce_found) {
+    b64(s->nonce_label, 32, nv);
+    std::fprintf(f, ",\n  \"Nonce\": %llu,\n  \"NonceValue\": \"%s\"",
+                 (unsigned long long)s->nonce_idx, nv.c_str());
+  }
+  std::fprintf(f, "\n}\n");
+  std::fclose(f);
+  return POST_OK;
+}
+
+int post_init_step(PostInitSession *s, uint64_t max_labels, uint64_t *done) {
+  if (!s || !done) return POST_ERR_INVALID_ARGS;
+  *done = 0;
+  int rc = require_gpu(s->cfg.provider_id);
+  if (rc != POST_OK) return rc;
+
+  LabelKernelArgs args;
+  std::memset(&args, 0, sizeof(args));
+  load_commitment_words(s->commitment, args.commitment_le);
+  args.scrypt_n = s->cfg.scrypt_n;
+  args.out_full = 0;
+  args.scratch = s->d_scratch;
+  args.scratch_lanes = s->lanes;
+  args.indices = nullptr;
+  args.commit_ids = nullptr;
+  args.commitments = nullptr;
+  args.has_difficulty = 1;
+  args.cand = s->d_cand;
+  args.cand_count = s->d_cand_count;
+  args.cand_cap = CAND_CAP;
+
+  uint8_t cur_difficulty[32];
+  std::memset(cur_difficulty, 0xff, 32); /* track the global minimum */
+  {
+    std::lock_guard<std::mutex> lk(s->nonce_mu);
+    if (s->nonce_found) std::memcpy(cur_difficulty, s->nonce_label, 32);
+  }
+
+  const uint32_t blocks = (uint32_t)(s->lanes / THREADS);
+  std::vector<PostVrfCandidate> cands(CAND_CAP);
+
+  if (!s->ev0) {
+    HIP_TRY(hipEventCreate(&s->ev0));
+    HIP_TRY(hipEventCreate(&s->ev1));
+  }
+  s->last_kernel_ms = 0;
+
+  uint64_t range = s->range_end - s->range_start;
+  uint64_t pos = s->written.load(); /* relative position (resume) */
+  uint64_t target = std::min(range, pos + max_labels);
+  while (pos < target) {
+    if (s->cancel.load()) return POST_ERR_CANCELLED;
+    uint64_t count = std::min(s->batch, target - pos);
+    args.start = s->range_start + pos;
+    args.count = count;
+    args.out = s->keep_all ? s->d_all + pos * POST_LABEL_SIZE : s->d_out;
+    difficulty_to_be_words(cur_difficulty, args.difficulty_be);
+    unsigned int zero = 0;
+    HIP_TRY(hipMemcpyAsync(s->d_cand_count, &zero, sizeof(zero),
+                           hipMemcpyHostToDevice, s->stream));
+    HIP_TRY(hipEventRecord(s->ev0, s->stream));
+    HIP_TRY(poste_launch_label_kernel(&args, blocks, s->stream));
+    HIP_TRY(hipEventRecord(s->ev1, s->stream));
+    HIP_TRY(hipMemcpyAsync(s->h_batch, args.out, count * POST_LABEL_SIZE,
+                           hipMemcpyDeviceToHost, s->stream));
+    unsigned int n_cand = 0;
+    HIP_TRY(hipMemcpyAsync(&n_cand, s->d_cand_count, sizeof(n_cand),
+                           hipMemcpyDeviceToHost, s->stream));
+    HIP_TRY(hipStreamSynchronize(s->stream));
+    if (n_cand > 0) {
+      unsigned int take = std::min(n_cand, CAND_CAP);
+      HIP_TRY(hipMemcpy(cands.data(), s->d_cand,
+                        sizeof(PostVrfCandidate) * take,
+                        hipMemcpyDeviceToHost));
+      std::lock_guard<std::mutex> lk(s->nonce_mu);
+      for (unsigned int i = 0; i < take; i++) {
+        uint8_t lab[32];
+        for (int k = 0; k < 8; k++) {
+          uint32_t w = cands[i].label_be[k];
+          lab[4 * k] = (uint8_t)(w >> 24);
+          lab[4 * k + 1] = (uint8_t)(w >> 16);
+          lab[4 * k + 2] = (uint8_t)(w >> 8);
+          lab[4 * k + 3] = (uint8_t)w;
+        }
+        int c = std::memcmp(lab, cur_difficulty, 32);
+        if (!s->nonce_found || c < 0 ||
+            (c == 0 && cands[i].index < s->nonce_idx)) {
+          s->nonce_found = true;
+          s->nonce_idx = cands[i].index;
+          std::memcpy(s->nonce_label, lab, 32);
+          std::memcpy(cur_difficulty, lab, 32);
+        }
+      }
+    }
+
+    /* reference-label self-check (ErrReferenceLabelMismatch semantics,
+     * activation/post.go:299-312): recompute one label of the batch on the
+     * host and compare */
+    {
+      uint64_t probe = count / 2;
+      uint8_t ref[32];
+      if (poste::host_label(s->commitment, args.start + probe,
+                            s->cfg.scrypt_n, ref) != 0) {
+        set_error("host reference label failed");
+        return POST_ERR;
+      }
+      if (std::memcmp(ref, s->h_batch + probe * POST_LABEL_SIZE,
+                      POST_LABEL_SIZE) != 0) {
+        set_error("reference label mismatch: device labels diverge from the "
+                  "host reference (ErrReferenceLabelMismatch)");
+        return POST_ERR;
+      }
+    }
+
+    if (!s->data_dir.empty()) {
+      rc = write_batch_files(s, s->range_start + pos, count, s->h_batch);
+      if (rc != POST_OK) return rc;
+    }
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, s->ev0, s->ev1);
+    s->last_kernel_ms += ms;
+    pos += count;
+    s->written.store(pos);
+    *done += count;
+  }
+  return POST_OK;
+}
+
+double post_init_last_kernel_ms(const PostInitSession *s) {
+  return s ? s->last_kernel_ms : 0.0;
+}
+
+int post_init_run(PostInitSession *s) {
+  if (!s) return POST_ERR_INVALID_ARGS;
+  uint64_t range = s->range_end - s->range_start;
+  while (s->written.load() < range) {
+    uint64_t done = 0;
+    int rc = post_init_step(s, s->batch * 8, &done);
+    if (rc != POST_OK) return rc;
+    if (done == 0) break;
+  }
+  if (!s->data_dir.empty()) return write_metadata(s);
+  return POST_OK;
+}
+
+int post_init_copy_labels(const PostInitSession *s, uint64_t first,
+                          uint64_t count, uint8_t *out_host) {
+  if (!s || !out_host) return POST_ERR_INVALID_ARGS;
+  if (!s->keep_all) {
+    set_error("labels were not kept on device (file mode or large range)");
+    return POST_ERR_INVALID_ARGS;
+  }
+  if (first + count > s->range_end - s->range_start) {
+    set_error("range out of bounds");
+    return POST_ERR_INVALID_ARGS;
+  }
+  HIP_TRY(hipMemcpy(out_host, s->d_all + first * POST_LABEL_SIZE,
+                    count * POST_LABEL_SIZE, hipMemcpyDeviceToHost));
+  return POST_OK;
+}
+
+void post_init_free(PostInitSession *s) { delete s; }
+
+/* ------------------------- benchmark ------------------------- */
+
+int post_benchmark(uint32_t provider_id, uint32_t scrypt_n,
+                   uint64_t *labels_per_sec) {
+  int rc = require_gpu(provider_id);
+  if (rc != POST_OK) return rc;
+  PostInitConfig cfg;
+  std::memset(&cfg, 0, sizeof(cfg));
+  std::memset(cfg.node_id, 0xA5, 32);
+  std::memset(cfg.commitment_atx_id, 0x5A, 32);
+  cfg.num_units = 1;
+  cfg.labels_per_unit = 1ull << 18;
+  cfg.max_file_size = 1ull << 40;
+  cfg.scrypt_n = scrypt_n;
+  cfg.provider_id = provider_id;
+  cfg.scratch_bytes = 8ull << 30;
+  PostInitSession *s = nullptr;
+  rc = post_init_new(&cfg, &s);
+  if (rc != POST_OK) return rc;
+  hipEvent_t t0, t1;
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, nullptr);
+  rc = post_init_run(s);
+  (void)hipEventRecord(t1, nullptr);
+  (void)hipEventSynchronize(t1);
+  float ms = 0;
+  (void)hipEventElapsedTime(&ms, t0, t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
+  if (rc == POST_OK && ms > 0)
+    *labels_per_sec = (uint64_t)((double)cfg.labels_per_unit / (ms / 1e3));
+  post_init_free(s);
+  return rc;
+}
+
+/* ------------------------- proving ------------------------- */
+
+static int prove_core(const uint8_t *host_labels, /* chunked source */
+                      uint64_t num_labels, const PostProveConfig *cfg,
+                      PostProof *out) {
+  if (cfg->nonces == 0 || cfg->nonces % POSTE_NONCE_GROUP != 0) {
+    set_error("nonces must be a positive multiple of 16");
+    return POST_ERR_INVALID_ARGS;
+  }
+  if (cfg->pow_mode != POST_POW_MODE_BLAKE3) {
+    set_error("RandomX k2pow is not supported by this engine (parity-"
+              "unpinned; use POST_POW_MODE_BLAKE3)");
+    return POST_ERR_UNSUPPORTED;
+  }
+  int rc = require_gpu(cfg->provider_id);
+  if (rc != POST_OK) return rc;
+  int dev = (int)cfg->provider_id;
+  DeviceTables tbl;
+  rc = get_aes_tables(dev, tbl);
+  if (rc != POST_OK) return rc;
+
+  const uint32_t n_ciphers = cfg->nonces / POSTE_NONCES_PER_AES;
+  const uint32_t n_groups = cfg->nonces / POSTE_NONCE_GROUP;
+  std::vector<uint64_t> group_pow(n_groups);
+  for (uint32_t g = 0; g < n_groups; g++)
+    group_pow[g] = poste::k2pow_search_blake3(cfg->challenge, g,
+                                              cfg->pow_difficulty,
+                                              cfg->pow_threads);
+  std::vector<uint32_t> rk((size_t)n_ciphers * 44);
+  for (uint32_t c = 0; c < n_ciphers; c++) {
+    uint8_t key[16];
+    uint32_t grp = (c * POSTE_NONCES_PER_AES) / POSTE_NONCE_GROUP;
+    poste::prove_cipher_key(cfg->challenge, c, group_pow[grp], key);
+    poste::aes128_expand(key, rk.data() + (size_t)c * 44);
+  }
+  uint32_t *d_rk = nullptr;
+  HIP_TRY(hipMalloc(&d_rk, rk.size() * 4));
+  HIP_TRY(hipMemcpy(d_rk, rk.data(), rk.size() * 4, hipMemcpyHostToDevice));
+
+  const uint64_t CHUNK = 1ull << 24; /* 16M labels = 256 MiB per chunk */
+  uint8_t *d_labels = nullptr;
+  HIP_TRY(hipMalloc(&d_labels, std::min(CHUNK, num_labels) * 16));
+  const uint32_t HIT_CAP = 1u << 22;
+  PostScanHit *d_hits = nullptr;
+  unsigned int *d_hit_count = nullptr;
+  HIP_TRY(hipMalloc(&d_hits, sizeof(PostScanHit) * HIT_CAP));
+  HIP_TRY(hipMalloc(&d_hit_count, 4));
+
+  ScanKernelArgs sa;
+  std::memset(&sa, 0, sizeof(sa));
+  sa.te = tbl.d_te;
+  sa.sbox = tbl.d_sbox;
+  sa.rk = d_rk;
+  sa.n_ciphers = n_ciphers;
+  sa.difficulty = poste::proving_difficulty(cfg->k1, num_labels);
+  sa.hits = d_hits;
+  sa.hit_count = d_hit_count;
+  sa.hit_cap = HIT_CAP;
+
+  std::vector<PostScanHit> all_hits;
+  for (uint64_t base = 0; base < num_labels; base += CHUNK) {
+    uint64_t cnt = std::min(CHUNK, num_labels - base);
+    HIP_TRY(hipMemcpy(d_labels, host_labels + base * 16, cnt * 16,
+                      hipMemcpyHostToDevice));
+    unsigned int zero = 0;
+    HIP_TRY(hipMemcpy(d_hit_count, &zero, 4, hipMemcpyHostToDevice));
+    sa.labels = (const uint4 *)d_labels;
+    sa.count = cnt;
+    sa.index_base = base;
+    uint32_t blocks = (uint32_t)std::min<uint64_t>(
+        (cnt + THREADS - 1) / THREADS, 8192);
+    HIP_TRY(poste_launch_scan_kernel(&sa, blocks, nullptr));
+    HIP_TRY(hipDeviceSynchronize());
+    unsigned int n_hits = 0;
+    HIP_TRY(hipMemcpy(&n_hits, d_hit_count, 4, hipMemcpyDeviceToHost));
+    if (n_hits > HIT_CAP) n_hits = HIT_CAP;
+    size_t off = all_hits.size();
+    all_hits.resize(off + n_hits);
+    if (n_hits)
+      HIP_TRY(hipMemcpy(all_hits.data() + off, d_hits,
+                        sizeof(PostScanHit) * n_hits, hipMemcpyDeviceToHost));
+  }
+  (void)hipFree(d_labels);
+  (void)hipFree(d_hits);
+  (void)hipFree(d_hit_count);
+  (void)hipFree(d_rk);
+
+  /* winner: nonce whose k2-th smallest passing index is smallest
+   * (streaming-order first across the ascending scan); tie -> lowest nonce */
+  std::vector<std::vector<uint64_t>> per_nonce(cfg->nonces);
+  for (const auto &h : all_hits) per_nonce[h.nonce].push_back(h.index);
+  int64_t best_nonce = -1;
+  uint64_t best_kth = UINT64_MAX;
+  for (uint32_t nn = 0; nn < cfg->nonces; nn++) {
+    auto &v = per_nonce[nn];
+    if (v.size() < cfg->k2) continue;
+    std::sort(v.begin(), v.end());
+    uint64_t kth = v[cfg->k2 - 1];
+    if (kth < best_kth) {
+      best_kth = kth;
+      best_nonce = nn;
+    }
+  }
+  if (best_nonce < 0) {
+    set_error("no nonce reached k2 passing indices");
+    return POST_ERR_NO_NONCE;
+  }
+  out->nonce = (uint32_t)best_nonce;
+  out->pow = group_pow[best_nonce / POSTE_NONCE_GROUP];
+  out->num_indices = (uint16_t)cfg->k2;
+  uint32_t bpi = poste::bits_per_index(num_labels);
+  out->indices_len = poste::pack_indices(per_nonce[best_nonce].data(),
+                                         cfg->k2, bpi, out->indices,
+                                         POST_MAX_INDICES_BYTES);
+  if (!out->indices_len) {
+    set_error("indices exceed the 800-byte wire cap");
+    return POST_ERR;
+  }
+  return POST_OK;
+}
+
+int post_prove_buffer(const uint8_t *labels, uint64_t num_labels,
+                      const uint8_t node_id[32],
+                      const uint8_t commitment_atx_id[32],
+                      const PostProveConfig *cfg, PostProof *out) {
+  (void)node_id;
+  (void)commitment_atx_id; /* scan operates on labels only */
+  if (!labels || !cfg || !out) return POST_ERR_INVALID_ARGS;
+  return prove_core(labels, num_labels, cfg, out);
+}
+
+int post_prove(const char *data_dir, const PostProveConfig *cfg,
+               PostProof *out) {
+  if (!data_dir || !cfg || !out) return POST_ERR_INVALID_ARGS;
+  /* concatenate postdata_*.bin in index order */
+  std::vector<uint8_t> labels;
+  for (uint64_t i = 0;; i++) {
+    char path[4096];
+    std::snprintf(path, sizeof path, "%s/postdata_%llu.bin", data_dir,
+                  (unsigned long long)i);
+    FILE *f = std::fopen(path, "rb");
+    if (!f) break;
+    std::fseek(f, 0, SEEK_END);
+    long sz = std::ftell(f);
+    std::fseek(f, 0, SEEK_SET);
+    size_t off = labels.size();
+    labels.resize(off + (size_t)sz);
+    if (std::fread(labels.data() + off, 1, (size_t)sz, f) != (size_t)sz) {
+      std::fclose(f);
+      set_error("short read");
+      return POST_ERR_IO;
+    }
+    std::fclose(f);
+  }
+  if (labels.empty()) {
+    set_error("no postdata_*.bin in data_dir");
+    return POST_ERR_IO;
+  }
+  return prove_core(labels.data(), labels.size() / 16, cfg, out);
+}
+
+/* ------------------------- verification ------------------------- */
+
+int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
+                      uint32_t n, const PostVerifyConfig *cfg, int *statuses,
+                      uint32_t *invalid_indices) {
+  if (!proofs || !metas || !cfg || !statuses) return POST_ERR_INVALID_ARGS;
+  if (cfg->pow_mode != POST_POW_MODE_BLAKE3) {
+    set_error("RandomX k2pow is not supported (use POST_POW_MODE_BLAKE3)");
+    return POST_ERR_UNSUPPORTED;
+  }
+  int rc = require_gpu(cfg->provider_id);
+  if (rc != POST_OK) return rc;
+
+  struct Task {
+    uint64_t label_index;
+    uint32_t proof;
+    uint32_t position; /* position within the proof's k2 indices */
+  };
+  std::vector<Task> tasks;
+  std::vector<uint32_t> commit_words; /* 8 words per proof */
+  std::vector<std::vector<uint64_t>> proof_indices(n);
+
+  for (uint32_t p = 0; p < n; p++) {
+    statuses[p] = POST_OK;
+    const PostProof &pr = proofs[p];
+    const PostProofMetadata &me = metas[p];
+    uint64_t num_labels = (uint64_t)me.num_units * me.labels_per_unit;
+    uint32_t bpi = poste::bits_per_index(num_labels);
+    if (pr.num_indices != cfg->k2 ||
+        pr.indices_len != ((uint64_t)cfg->k2 * bpi + 7) / 8) {
+      statuses[p] = POST_ERR_INVALID_ARGS;
+      continue;
+    }
+    uint32_t group = pr.nonce / POSTE_NONCE_GROUP;
+    if (poste::k2pow_verify_blake3(me.challenge, group, pr.pow,
+                                   cfg->pow_difficulty) != 0) {
+      statuses[p] = POST_ERR_POW;
+      continue;
+    }
+    proof_indices[p].resize(cfg->k2);
+    poste::unpack_indices(pr.indices, cfg->k2, bpi,
+                          proof_indices[p].data());
+    uint8_t cm[32];
+    poste::commitment(me.node_id, me.commitment_atx_id, cm);
+    uint32_t cw[8];
+    std::memcpy(cw, cm, 32);
+    for (int k = 0; k < 8; k++) commit_words.push_back(cw[k]);
+
+    std::vector<uint32_t> positions;
+    if (cfg->selected_index >= 0) {
+      positions.push_back((uint32_t)cfg->selected_index);
+    } else if (cfg->subset_seed && cfg->k3 < cfg->k2) {
+      positions.resize(cfg->k3);
+      poste::subset_positions(cfg->k2, cfg->k3, cfg->subset_seed,
+                              cfg->subset_seed_len, positions.data());
+    } else {
+      positions.resize(cfg->k2);
+      for (uint32_t i = 0; i < cfg->k2; i++) positions[i] = i;
+    }
+    for (uint32_t pos : positions) {
+      if (pos >= cfg->k2) {
+        statuses[p] = POST_ERR_INVALID_ARGS;
+        break;
+      }
+      uint64_t li = proof_indices[p][pos];
+      if (li >= num_labels) {
+        statuses[p] = POST_ERR_INVALID_INDEX;
+        if (invalid_indices) invalid_indices[p] = pos;
+        break;
+      }
+      tasks.push_back({li, p, pos});
+    }
+  }
+
+  /* drop tasks of proofs already failed */
+  tasks.erase(std::remove_if(tasks.begin(), tasks.end(),
+                             [&](const Task &t) {
+                               return statuses[t.proof] != POST_OK;
+                             }),
+              tasks.end());
+  if (tasks.empty()) return POST_OK;
+
+  /* scratch-bounded chunks of label recomputes on the GPU */
+  size_t free_b = 0, total_b = 0;
+  HIP_TRY(hipMemGetInfo(&free_b, &total_b));
+  uint64_t per_lane = (uint64_t)cfg->scrypt_n * 128;
+  uint64_t max_lanes = (uint64_t)((double)free_b * 0.75) / per_lane;
+  max_lanes = std::min<uint64_t>(max_lanes, 512ull * 1024);
+  max_lanes = (max_lanes / THREADS) * THREADS;
+  if (max_lanes == 0) {
+    set_error("not enough memory for verification scratch");
+    return POST_ERR_OOM;
+  }
+
+  std::vector<uint64_t> h_idx(tasks.size());
+  std::vector<uint32_t> h_cid(tasks.size());
+  for (size_t i = 0; i < tasks.size(); i++) {
+    h_idx[i] = tasks[i].label_index;
+    h_cid[i] = tasks[i].proof;
+  }
+  uint64_t lanes = std::min<uint64_t>(max_lanes,
+                                      ((tasks.size() + THREADS - 1) / THREADS) *
+                                          THREADS);
+  uint32_t *d_scratch = nullptr;
+  uint64_t *d_idx = nullptr;
+  uint32_t *d_cid = nullptr, *d_cm = nullptr;
+  uint8_t *d_out = nullptr;
+  HIP_TRY(hipMalloc(&d_scratch, (size_t)lanes * per_lane));
+  HIP_TRY(hipMalloc(&d_idx, h_idx.size() * 8));
+  HIP_TRY(hipMalloc(&d_cid, h_cid.size() * 4));
+  HIP_TRY(hipMalloc(&d_cm, commit_words.size() * 4));
+  HIP_TRY(hipMalloc(&d_out, tasks.size() * 32));
+  HIP_TRY(hipMemcpy(d_idx, h_idx.data(), h_idx.size() * 8,
+                    hipMemcpyHostToDevice));
+  HIP_TRY(hipMemcpy(d_cid, h_cid.data(), h_cid.size() * 4,
+                    hipMemcpyHostToDevice));
+  HIP_TRY(hipMemcpy(d_cm, commit_words.data(), commit_words.size() * 4,
+                    hipMemcpyHostToDevice));
+
+  LabelKernelArgs la;
+  std::memset(&la, 0, sizeof(la));
+  la.scrypt_n = cfg->scrypt_n;
+  la.out_full = 1;
+  la.scratch = d_scratch;
+  la.scratch_lanes = lanes;
+  la.out = d_out;
+  la.indices = h_idx.size() ? d_idx : nullptr;
+  la.commit_ids = d_cid;
+  la.commitments = d_cm;
+  la.count = tasks.size();
+  HIP_TRY(poste_launch_label_kernel(&la, (uint32_t)(lanes / THREADS),
+                                    nullptr));
+  HIP_TRY(hipDeviceSynchronize());
+  std::vector<uint8_t> full((size_t)tasks.size() * 32);
+  HIP_TRY(hipMemcpy(full.data(), d_out, full.size(),
+                    hipMemcpyDeviceToHost));
+  (void)hipFree(d_scratch);
+  (void)hipFree(d_idx);
+  (void)hipFree(d_cid);
+  (void)hipFree(d_cm);
+  (void)hipFree(d_out);
+
+  /* final AES threshold predicate on host (cheap; K3*n blocks) */
+  std::vector<std::array<uint32_t, 44>> rk_cache(n);
+  std::vector<bool> rk_ready(n, false);
+  for (size_t i = 0; i < tasks.size(); i++) {
+    const Task &t = tasks[i];
+    if (statuses[t.proof] != POST_OK) continue;
+    const PostProof &pr = proofs[t.proof];
+    const PostProofMetadata &me = metas[t.proof];
+    if (!rk_ready[t.proof]) {
+      uint8_t key[16];
+      poste::prove_cipher_key(me.challenge,
+                              pr.nonce / POSTE_NONCES_PER_AES, pr.pow, key);
+      poste::aes128_expand(key, rk_cache[t.proof].data());
+      rk_ready[t.proof] = true;
+    }
+    uint8_t out16[16];
+    poste::aes128_enc_block(rk_cache[t.proof].data(), &full[i * 32], out16);
+    uint32_t half = pr.nonce % POSTE_NONCES_PER_AES;
+    uint64_t v = 0;
+    for (int b = 0; b < 8; b++)
+      v |= (uint64_t)out16[8 * half + b] << (8 * b);
+    uint64_t num_labels = (uint64_t)me.num_units * me.labels_per_unit;
+    if (v >= poste::proving_difficulty(cfg->k1, num_labels)) {
+      statuses[t.proof] = POST_ERR_INVALID_INDEX;
+      if (invalid_indices) invalid_indices[t.proof] = t.position;
+    }
+  }
+  return POST_OK;
+}
+
+int post_verify(const PostProof *proof, const PostProofMetadata *meta,
+                const PostVerifyConfig *cfg, uint32_t *invalid_index) {
+  int status = 0;
+  uint32_t inv = 0;
+  int rc = post_verify_batch(proof, meta, 1, cfg, &status, &inv);
+  if (rc != POST_OK) return rc;
+  if (invalid_index) *invalid_index = inv;
+  return status;
+}
+
+int post_verify_vrf_nonce(const PostProofMetadata *meta, uint64_t index,
+                          uint32_t scrypt_n, uint32_t provider_id) {
+  if (!meta) return POST_ERR_INVALID_ARGS;
+  int rc = require_gpu(provider_id);
+  if (rc != POST_OK) return rc;
+  uint64_t num_labels = (uint64_t)meta->num_units * meta->labels_per_unit;
+  uint8_t cm[32];
+  poste::commitment(meta->node_id, meta->commitment_atx_id, cm);
+
+  uint64_t per_lane = (uint64_t)scrypt_n * 128;
+  uint32_t *d_scratch = nullptr;
+  uint64_t *d_idx = nullptr;
+  uint8_t *d_out = nullptr;
+  HIP_TRY(hipMalloc(&d_scratch, (size_t)THREADS * per_lane));
+  HIP_TRY(hipMalloc(&d_idx, 8));
+  HIP_TRY(hipMalloc(&d_out, 32));
+  HIP_TRY(hipMemcpy(d_idx, &index, 8, hipMemcpyHostToDevice));
+  LabelKernelArgs la;
+  std::memset(&la, 0, sizeof(la));
+  load_commitment_words(cm, la.commitment_le);
+  la.scrypt_n = scrypt_n;
+  la.out_full = 1;
+  la.scratch = d_scratch;
+  la.scratch_lanes = THREADS;
+  la.out = d_out;
+  la.indices = d_idx;
+  la.count = 1;
+  HIP_TRY(poste_launch_label_kernel(&la, 1, nullptr));
+  HIP_TRY(hipDeviceSynchronize());
+  uint8_t full[32];
+  HIP_TRY(hipMemcpy(full, d_out, 32, hipMemcpyDeviceToHost));
+  (void)hipFree(d_scratch);
+  (void)hipFree(d_idx);
+  (void)hipFree(d_out);
+  uint8_t difficulty[32];
+  poste::vrf_difficulty(num_labels, difficulty);
+  if (std::memcmp(full, difficulty, 32) < 0) return POST_OK;
+  set_error("vrf nonce label above threshold");
+  return POST_ERR_INVALID_INDEX;
+}
+
+/* ------------------------- self-test exports ------------------------- */
+
+void post_selftest_blake3(const uint8_t *msg, size_t len, uint8_t out[32]) {
+  poste::blake3(msg, len, out, 32);
+}
+
+void post_selftest_aes128(const uint8_t key[16], const uint8_t in[16],
+                          uint8_t out[16]) {
+  uint32_t rk[44];
+  poste::aes128_expand(key, rk);
+  poste::aes128_enc_block(rk, in, out);
+}
+
+int post_selftest_label(const uint8_t node_id[32],
+                        const uint8_t commitment_atx_id[32], uint64_t index,
+                        uint32_t scrypt_n, uint8_t out[32]) {
+  uint8_t cm[32];
+  poste::commitment(node_id, commitment_atx_id, cm);
+  return poste::host_label(cm, index, scrypt_n, out);
+}
+
+} /* extern "C" */

@@ -1,0 +1,50 @@
+/* kernel_args.h — argument blocks shared between engine.cpp (host) and
+ * kernels.hip (device).  Kept POD; pointers are device pointers. */
+#ifndef POST_KERNEL_ARGS_H
+#define POST_KERNEL_ARGS_H
+
+#include <hip/hip_runtime.h>
+
+#include "post_common.h"
+
+struct LabelKernelArgs {
+  uint32_t commitment_le[8];
+  uint64_t start;
+  uint64_t count;
+  uint32_t scrypt_n;
+  uint32_t out_full;
+  uint32_t *scratch;
+  uint64_t scratch_lanes;
+  uint8_t *out;
+  const uint64_t *indices;
+  const uint32_t *commit_ids;
+  const uint32_t *commitments;
+  uint32_t has_difficulty;
+  uint32_t difficulty_be[8];
+  PostVrfCandidate *cand;
+  unsigned int *cand_count;
+  uint32_t cand_cap;
+};
+
+struct ScanKernelArgs {
+  const uint4 *labels;
+  uint64_t count;
+  uint64_t index_base;
+  const uint32_t *te;
+  const uint8_t *sbox;
+  const uint32_t *rk;
+  uint32_t n_ciphers;
+  uint64_t difficulty;
+  PostScanHit *hits;
+  unsigned int *hit_count;
+  uint32_t hit_cap;
+};
+
+extern "C" {
+hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
+                                     uint32_t blocks, hipStream_t stream);
+hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
+                                    uint32_t blocks, hipStream_t stream);
+}
+
+#endif

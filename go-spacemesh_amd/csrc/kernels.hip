@@ -1,0 +1,463 @@
+/* kernels.hip — MI355X (gfx950/CDNA4) kernels for the POST hot path.
+ *
+ * post_label_kernel: scrypt-N labeling (the init hot loop the reference
+ *   reaches at activation/post.go:295).  One lane = one label: the ROMix
+ *   working block X (2x64 B) lives in VGPRs, the N-entry scratchpad V
+ *   (128 B * N per in-flight label; 1 MiB at mainnet N=8192) lives in HBM,
+ *   block-interleaved across lanes at 128-B granularity so each lane's
+ *   per-iteration access is one fully-used 128-B burst (phase-1 stores merge
+ *   to full lines in the XCD L2; phase-2 reads are random 128-B gathers —
+ *   the HBM-traffic roofline of SURVEY.md §8(d): ~2 MiB/label).
+ *   No MFMA: this is a u32 add/xor/rotate hash loop (BASELINE.json
+ *   north_star), VALU ~1.36e7 int ops/label.
+ *   Also used in index-list mode for verification's label recompute
+ *   (validation.go:182-222 -> K3 sampled indices) with per-task commitments.
+ *
+ * post_scan_kernel: the proving index scan (post-service GenProof,
+ *   api/grpcserver/post_client.go:69-143): AES-128 over each 16-B label for
+ *   n_ciphers ciphers (2 nonces per cipher), T-tables + round keys staged in
+ *   LDS, passing (nonce,index) pairs appended with a global atomic.
+ *
+ * Wave size 64, 256-thread workgroups.  Grid-stride loops size the in-flight
+ * label set to the scratch allocation, independent of batch size.
+ */
+#include <hip/hip_runtime.h>
+
+#include "kernel_args.h"
+#include "post_common.h"
+
+#define POSTE_THREADS 256
+
+/* ------------------------- SHA-256 (device) ------------------------- */
+__constant__ uint32_t c_sha_k[64] = {
+    0x428a2f98, 0x71374491, 0xb5c0fbcf, 0xe9b5dba5, 0x3956c25b, 0x59f111f1,
+    0x923f82a4, 0xab1c5ed5, 0xd807aa98, 0x12835b01, 0x243185be, 0x550c7dc3,
+    0x72be5d74, 0x80deb1fe, 0x9bdc06a7, 0xc19bf174, 0xe49b69c1, 0xefbe4786,
+    0x0fc19dc6, 0x240ca1cc, 0x2de92c6f, 0x4a7484aa, 0x5cb0a9dc, 0x76f988da,
+    0x983e5152, 0xa831c66d, 0xb00327c8, 0xbf597fc7, 0xc6e00bf3, 0xd5a79147,
+    0x06ca6351, 0x14292967, 0x27b70a85, 0x2e1b2138, 0x4d2c6dfc, 0x53380d13,
+    0x650a7354, 0x766a0abb, 0x81c2c92e, 0x92722c85, 0xa2bfe8a1, 0xa81a664b,
+    0xc24b8b70, 0xc76c51a3, 0xd192e819, 0xd6990624, 0xf40e3585, 0x106aa070,
+    0x19a4c116, 0x1e376c08, 0x2748774c, 0x34b0bcb5, 0x391c0cb3, 0x4ed8aa4a,
+    0x5b9cca4f, 0x682e6ff3, 0x748f82ee, 0x78a5636f, 0x84c87814, 0x8cc70208,
+    0x90befffa, 0xa4506ceb, 0xbef9a3f7, 0xc67178f2};
+
+__device__ __forceinline__ uint32_t ror32(uint32_t x, int n) {
+  return __builtin_rotateright32(x, n);
+}
+
+/* m holds big-endian message words; h updated in place. */
+__device__ void sha_compress(uint32_t h[8], const uint32_t m[16]) {
+  uint32_t w[16];
+#pragma unroll
+  for (int i = 0; i < 16; i++) w[i] = m[i];
+  uint32_t a = h[0], b = h[1], c = h[2], d = h[3];
+  uint32_t e = h[4], f = h[5], g = h[6], hh = h[7];
+#pragma unroll
+  for (int t = 0; t < 64; t++) {
+    uint32_t wt;
+    if (t < 16) {
+      wt = w[t];
+    } else {
+      uint32_t a15 = w[(t - 15) & 15], a2 = w[(t - 2) & 15];
+      wt = w[t & 15] += (ror32(a15, 7) ^ ror32(a15, 18) ^ (a15 >> 3)) +
+                        w[(t - 7) & 15] +
+                        (ror32(a2, 17) ^ ror32(a2, 19) ^ (a2 >> 10));
+    }
+    uint32_t t1 = hh + (ror32(e, 6) ^ ror32(e, 11) ^ ror32(e, 25)) +
+                  ((e & f) ^ (~e & g)) + c_sha_k[t] + wt;
+    uint32_t t2 = (ror32(a, 2) ^ ror32(a, 13) ^ ror32(a, 22)) +
+                  ((a & b) ^ (a & c) ^ (b & c));
+    hh = g; g = f; f = e; e = d + t1;
+    d = c; c = b; b = a; a = t1 + t2;
+  }
+  h[0] += a; h[1] += b; h[2] += c; h[3] += d;
+  h[4] += e; h[5] += f; h[6] += g; h[7] += hh;
+}
+
+__device__ __forceinline__ void sha_init(uint32_t h[8]) {
+  h[0] = 0x6a09e667; h[1] = 0xbb67ae85; h[2] = 0x3c6ef372; h[3] = 0xa54ff53a;
+  h[4] = 0x510e527f; h[5] = 0x9b05688c; h[6] = 0x1f83d9ab; h[7] = 0x5be0cd19;
+}
+
+/* ------------------------- salsa20/8 + BlockMix ------------------------- */
+#define SALSA_QR(a, b, c, d)                                                   \
+  do {                                                                         \
+    x##b ^= __builtin_rotateleft32(x##a + x##d, 7);                            \
+    x##c ^= __builtin_rotateleft32(x##b + x##a, 9);                            \
+    x##d ^= __builtin_rotateleft32(x##c + x##b, 13);                           \
+    x##a ^= __builtin_rotateleft32(x##d + x##c, 18);                           \
+  } while (0)
+
+__device__ __forceinline__ void salsa8(uint32_t b[16]) {
+  uint32_t x0 = b[0], x1 = b[1], x2 = b[2], x3 = b[3], x4 = b[4], x5 = b[5],
+           x6 = b[6], x7 = b[7], x8 = b[8], x9 = b[9], x10 = b[10],
+           x11 = b[11], x12 = b[12], x13 = b[13], x14 = b[14], x15 = b[15];
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    SALSA_QR(0, 4, 8, 12);
+    SALSA_QR(5, 9, 13, 1);
+    SALSA_QR(10, 14, 2, 6);
+    SALSA_QR(15, 3, 7, 11);
+    SALSA_QR(0, 1, 2, 3);
+    SALSA_QR(5, 6, 7, 4);
+    SALSA_QR(10, 11, 8, 9);
+    SALSA_QR(15, 12, 13, 14);
+  }
+  b[0] += x0; b[1] += x1; b[2] += x2; b[3] += x3;
+  b[4] += x4; b[5] += x5; b[6] += x6; b[7] += x7;
+  b[8] += x8; b[9] += x9; b[10] += x10; b[11] += x11;
+  b[12] += x12; b[13] += x13; b[14] += x14; b[15] += x15;
+}
+
+/* BlockMix for r=1: (B0,B1) -> (Salsa(B0^B1), Salsa(Salsa(B0^B1)^B1)) */
+__device__ __forceinline__ void blockmix_r1(uint32_t X[32]) {
+  uint32_t T[16];
+#pragma unroll
+  for (int k = 0; k < 16; k++) T[k] = X[k] ^ X[16 + k];
+  salsa8(T);
+#pragma unroll
+  for (int k = 0; k < 16; k++) {
+    uint32_t y0 = T[k];
+    T[k] = y0 ^ X[16 + k];
+    X[k] = y0;
+  }
+  salsa8(T);
+#pragma unroll
+  for (int k = 0; k < 16; k++) X[16 + k] = T[k];
+}
+
+/* ------------------------- label kernel ------------------------- */
+/* LabelKernelArgs: see kernel_args.h */
+
+/* PBKDF2 helper: one outer HMAC finalisation of a 32-byte inner digest */
+__device__ __forceinline__ void hmac_outer(const uint32_t ho[8],
+                                           const uint32_t inner[8],
+                                           uint32_t out[8]) {
+  uint32_t h[8], m[16];
+#pragma unroll
+  for (int i = 0; i < 8; i++) h[i] = ho[i];
+#pragma unroll
+  for (int i = 0; i < 8; i++) m[i] = inner[i];
+  m[8] = 0x80000000u;
+#pragma unroll
+  for (int i = 9; i < 15; i++) m[i] = 0;
+  m[15] = (64 + 32) * 8;
+  sha_compress(h, m);
+#pragma unroll
+  for (int i = 0; i < 8; i++) out[i] = h[i];
+}
+
+__global__ void __launch_bounds__(POSTE_THREADS)
+post_label_kernel(LabelKernelArgs a) {
+  const unsigned long long lane =
+      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const uint32_t n = a.scrypt_n;
+  const uint32_t mask = n - 1;
+  uint4 *V = (uint4 *)a.scratch;
+
+  /* per-lane running VRF minimum across this lane's tasks; reduced once per
+   * workgroup at kernel end so the candidate buffer stays bounded and the
+   * tracked minimum is exact (the initializer's nonce search,
+   * activation/post.go:295, keeps the global argmin label) */
+  uint32_t min_lab[8];
+  unsigned long long min_idx = 0;
+  int min_found = 0;
+
+  for (unsigned long long task = lane; task < a.count;
+       task += a.scratch_lanes) {
+    const unsigned long long index = a.indices ? a.indices[task]
+                                               : a.start + task;
+    const uint32_t *cw = a.commit_ids
+                             ? a.commitments + 8ull * a.commit_ids[task]
+                             : a.commitment_le;
+    /* password = commitment(32) || LE64(index): 10 LE words -> BE msg words */
+    uint32_t pw_be[10];
+#pragma unroll
+    for (int i = 0; i < 8; i++) pw_be[i] = __builtin_bswap32(cw[i]);
+    pw_be[8] = __builtin_bswap32((uint32_t)index);
+    pw_be[9] = __builtin_bswap32((uint32_t)(index >> 32));
+
+    /* HMAC ipad/opad chaining states for this password */
+    uint32_t hi[8], ho[8], m[16];
+    sha_init(hi);
+#pragma unroll
+    for (int i = 0; i < 10; i++) m[i] = pw_be[i] ^ 0x36363636u;
+#pragma unroll
+    for (int i = 10; i < 16; i++) m[i] = 0x36363636u;
+    sha_compress(hi, m);
+    sha_init(ho);
+#pragma unroll
+    for (int i = 0; i < 10; i++) m[i] = pw_be[i] ^ 0x5c5c5c5cu;
+#pragma unroll
+    for (int i = 10; i < 16; i++) m[i] = 0x5c5c5c5cu;
+    sha_compress(ho, m);
+
+    /* B = PBKDF2(P, "", 1, 128) -> X[32] little-endian salsa words */
+    uint32_t X[32];
+#pragma unroll
+    for (uint32_t blk = 1; blk <= 4; blk++) {
+      uint32_t h[8];
+#pragma unroll
+      for (int i = 0; i < 8; i++) h[i] = hi[i];
+      m[0] = blk;
+      m[1] = 0x80000000u;
+#pragma unroll
+      for (int i = 2; i < 15; i++) m[i] = 0;
+      m[15] = (64 + 4) * 8;
+      sha_compress(h, m);
+      uint32_t d[8];
+      hmac_outer(ho, h, d);
+#pragma unroll
+      for (int k = 0; k < 8; k++)
+        X[8 * (blk - 1) + k] = __builtin_bswap32(d[k]);
+    }
+
+    /* ROMix phase 1: V_j = X; X = BlockMix(X).  Scratch layout: the 128-B
+     * block of iteration j for this lane lives at uint4 offset
+     * (j*lanes + lane)*8 — per-lane bursts stay 128-B contiguous. */
+    {
+      unsigned long long base = lane * 8ull;
+      const unsigned long long stride = a.scratch_lanes * 8ull;
+      for (uint32_t j = 0; j < n; j++) {
+        uint4 *p = V + base;
+#pragma unroll
+        for (int c = 0; c < 8; c++)
+          p[c] = make_uint4(X[4 * c], X[4 * c + 1], X[4 * c + 2],
+                            X[4 * c + 3]);
+        blockmix_r1(X);
+        base += stride;
+      }
+    }
+    /* phase 2: j = Integerify(X) & (n-1); X ^= V_j; BlockMix */
+    for (uint32_t i = 0; i < n; i++) {
+      uint32_t j = X[16] & mask; /* low word suffices: n <= 2^32 */
+      const uint4 *p = V + ((unsigned long long)j * a.scratch_lanes + lane) *
+                               8ull;
+#pragma unroll
+      for (int c = 0; c < 8; c++) {
+        uint4 v = p[c];
+        X[4 * c] ^= v.x;
+        X[4 * c + 1] ^= v.y;
+        X[4 * c + 2] ^= v.z;
+        X[4 * c + 3] ^= v.w;
+      }
+      blockmix_r1(X);
+    }
+
+    /* out = PBKDF2(P, X-bytes, 1, 32) */
+    uint32_t h[8];
+#pragma unroll
+    for (int i = 0; i < 8; i++) h[i] = hi[i];
+#pragma unroll
+    for (int i = 0; i < 16; i++) m[i] = __builtin_bswap32(X[i]);
+    sha_compress(h, m);
+#pragma unroll
+    for (int i = 0; i < 16; i++) m[i] = __builtin_bswap32(X[16 + i]);
+    sha_compress(h, m);
+    m[0] = 1;
+    m[1] = 0x80000000u;
+#pragma unroll
+    for (int i = 2; i < 15; i++) m[i] = 0;
+    m[15] = (64 + 128 + 4) * 8;
+    sha_compress(h, m);
+    uint32_t lab_be[8]; /* full label as big-endian words */
+    hmac_outer(ho, h, lab_be);
+
+    if (a.out) {
+      if (a.out_full) {
+        uint4 *o = (uint4 *)(a.out + task * 32ull);
+        o[0] = make_uint4(__builtin_bswap32(lab_be[0]),
+                          __builtin_bswap32(lab_be[1]),
+                          __builtin_bswap32(lab_be[2]),
+                          __builtin_bswap32(lab_be[3]));
+        o[1] = make_uint4(__builtin_bswap32(lab_be[4]),
+                          __builtin_bswap32(lab_be[5]),
+                          __builtin_bswap32(lab_be[6]),
+                          __builtin_bswap32(lab_be[7]));
+      } else {
+        *(uint4 *)(a.out + task * 16ull) =
+            make_uint4(__builtin_bswap32(lab_be[0]),
+                       __builtin_bswap32(lab_be[1]),
+                       __builtin_bswap32(lab_be[2]),
+                       __builtin_bswap32(lab_be[3]));
+      }
+    }
+
+    if (a.has_difficulty) {
+      /* lexicographic byte compare == numeric compare of BE word sequence */
+      bool below_diff = false, below_min = false;
+#pragma unroll
+      for (int k = 0; k < 8; k++) {
+        if (lab_be[k] != a.difficulty_be[k]) {
+          below_diff = lab_be[k] < a.difficulty_be[k];
+          break;
+        }
+      }
+      if (below_diff) {
+        if (!min_found) {
+          below_min = true;
+        } else {
+          below_min = false;
+#pragma unroll
+          for (int k = 0; k < 8; k++) {
+            if (lab_be[k] != min_lab[k]) {
+              below_min = lab_be[k] < min_lab[k];
+              break;
+            }
+          }
+        }
+        if (below_min) {
+          min_found = 1;
+          min_idx = index;
+#pragma unroll
+          for (int k = 0; k < 8; k++) min_lab[k] = lab_be[k];
+        }
+      }
+    }
+  }
+
+  /* one exact candidate per workgroup: LDS gather + lane-0 scan */
+  if (a.has_difficulty) {
+    __shared__ uint32_t s_lab[POSTE_THREADS][8];
+    __shared__ unsigned long long s_idx[POSTE_THREADS];
+    __shared__ uint8_t s_found[POSTE_THREADS];
+    s_found[threadIdx.x] = (uint8_t)min_found;
+    if (min_found) {
+      s_idx[threadIdx.x] = min_idx;
+#pragma unroll
+      for (int k = 0; k < 8; k++) s_lab[threadIdx.x][k] = min_lab[k];
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      int best = -1;
+      for (int t = 0; t < POSTE_THREADS; t++) {
+        if (!s_found[t]) continue;
+        if (best < 0) {
+          best = t;
+          continue;
+        }
+        /* order by (label bytes, index) ascending */
+        int cmp = 0; /* -1: t < best, 0: equal labels, 1: t > best */
+        for (int k = 0; k < 8 && cmp == 0; k++) {
+          if (s_lab[t][k] < s_lab[best][k]) cmp = -1;
+          else if (s_lab[t][k] > s_lab[best][k]) cmp = 1;
+        }
+        if (cmp < 0 || (cmp == 0 && s_idx[t] < s_idx[best])) best = t;
+      }
+      if (best >= 0) {
+        unsigned int slot = atomicAdd(a.cand_count, 1u);
+        if (slot < a.cand_cap) {
+          a.cand[slot].index = s_idx[best];
+          for (int k = 0; k < 8; k++)
+            a.cand[slot].label_be[k] = s_lab[best][k];
+        }
+      }
+    }
+  }
+}
+
+/* ------------------------- proving scan kernel ------------------------- */
+/* ScanKernelArgs: see kernel_args.h */
+
+__global__ void __launch_bounds__(POSTE_THREADS)
+post_scan_kernel(ScanKernelArgs a) {
+  extern __shared__ uint32_t lds[];
+  uint32_t *sTe = lds;              /* 1024 */
+  uint32_t *sRk = lds + 1024;       /* n_ciphers*44 */
+  uint8_t *sSbox = (uint8_t *)(sRk + a.n_ciphers * 44); /* 256 */
+
+  for (uint32_t i = threadIdx.x; i < 1024; i += blockDim.x) sTe[i] = a.te[i];
+  for (uint32_t i = threadIdx.x; i < a.n_ciphers * 44; i += blockDim.x)
+    sRk[i] = a.rk[i];
+  for (uint32_t i = threadIdx.x; i < 256; i += blockDim.x)
+    sSbox[i] = a.sbox[i];
+  __syncthreads();
+
+  const unsigned long long stride =
+      (unsigned long long)gridDim.x * blockDim.x;
+  for (unsigned long long t =
+           (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+       t < a.count; t += stride) {
+    uint4 lraw = a.labels[t];
+    const uint32_t p0 = __builtin_bswap32(lraw.x),
+                   p1 = __builtin_bswap32(lraw.y),
+                   p2 = __builtin_bswap32(lraw.z),
+                   p3 = __builtin_bswap32(lraw.w);
+    for (uint32_t c = 0; c < a.n_ciphers; c++) {
+      const uint32_t *rk = sRk + c * 44;
+      uint32_t w0 = p0 ^ rk[0], w1 = p1 ^ rk[1], w2 = p2 ^ rk[2],
+               w3 = p3 ^ rk[3];
+#pragma unroll
+      for (int r = 1; r < 10; r++) {
+        uint32_t n0 = sTe[w0 >> 24] ^ sTe[256 + ((w1 >> 16) & 0xff)] ^
+                      sTe[512 + ((w2 >> 8) & 0xff)] ^ sTe[768 + (w3 & 0xff)] ^
+                      rk[4 * r];
+        uint32_t n1 = sTe[w1 >> 24] ^ sTe[256 + ((w2 >> 16) & 0xff)] ^
+                      sTe[512 + ((w3 >> 8) & 0xff)] ^ sTe[768 + (w0 & 0xff)] ^
+                      rk[4 * r + 1];
+        uint32_t n2 = sTe[w2 >> 24] ^ sTe[256 + ((w3 >> 16) & 0xff)] ^
+                      sTe[512 + ((w0 >> 8) & 0xff)] ^ sTe[768 + (w1 & 0xff)] ^
+                      rk[4 * r + 2];
+        uint32_t n3 = sTe[w3 >> 24] ^ sTe[256 + ((w0 >> 16) & 0xff)] ^
+                      sTe[512 + ((w1 >> 8) & 0xff)] ^ sTe[768 + (w2 & 0xff)] ^
+                      rk[4 * r + 3];
+        w0 = n0; w1 = n1; w2 = n2; w3 = n3;
+      }
+      uint32_t f0 = (((uint32_t)sSbox[w0 >> 24] << 24) |
+                     ((uint32_t)sSbox[(w1 >> 16) & 0xff] << 16) |
+                     ((uint32_t)sSbox[(w2 >> 8) & 0xff] << 8) |
+                     sSbox[w3 & 0xff]) ^ rk[40];
+      uint32_t f1 = (((uint32_t)sSbox[w1 >> 24] << 24) |
+                     ((uint32_t)sSbox[(w2 >> 16) & 0xff] << 16) |
+                     ((uint32_t)sSbox[(w3 >> 8) & 0xff] << 8) |
+                     sSbox[w0 & 0xff]) ^ rk[41];
+      uint32_t f2 = (((uint32_t)sSbox[w2 >> 24] << 24) |
+                     ((uint32_t)sSbox[(w3 >> 16) & 0xff] << 16) |
+                     ((uint32_t)sSbox[(w0 >> 8) & 0xff] << 8) |
+                     sSbox[w1 & 0xff]) ^ rk[42];
+      uint32_t f3 = (((uint32_t)sSbox[w3 >> 24] << 24) |
+                     ((uint32_t)sSbox[(w0 >> 16) & 0xff] << 16) |
+                     ((uint32_t)sSbox[(w1 >> 8) & 0xff] << 8) |
+                     sSbox[w2 & 0xff]) ^ rk[43];
+      /* per-nonce values: LE u64 of output bytes [0..8) and [8..16) */
+      unsigned long long v0 =
+          __builtin_bswap64(((unsigned long long)f0 << 32) | f1);
+      unsigned long long v1 =
+          __builtin_bswap64(((unsigned long long)f2 << 32) | f3);
+      if (v0 < a.difficulty) {
+        unsigned int s = atomicAdd(a.hit_count, 1u);
+        if (s < a.hit_cap) {
+          a.hits[s].index = a.index_base + t;
+          a.hits[s].nonce = c * POSTE_NONCES_PER_AES;
+        }
+      }
+      if (v1 < a.difficulty) {
+        unsigned int s = atomicAdd(a.hit_count, 1u);
+        if (s < a.hit_cap) {
+          a.hits[s].index = a.index_base + t;
+          a.hits[s].nonce = c * POSTE_NONCES_PER_AES + 1;
+        }
+      }
+    }
+  }
+}
+
+/* host-visible launchers (called from engine.cpp) */
+extern "C" {
+
+hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
+                                     uint32_t blocks, hipStream_t stream) {
+  hipLaunchKernelGGL(post_label_kernel, dim3(blocks), dim3(POSTE_THREADS), 0,
+                     stream, *args);
+  return hipGetLastError();
+}
+
+hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
+                                    uint32_t blocks, hipStream_t stream) {
+  size_t lds = 1024 * 4 + (size_t)args->n_ciphers * 44 * 4 + 256;
+  hipLaunchKernelGGL(post_scan_kernel, dim3(blocks), dim3(POSTE_THREADS), lds,
+                     stream, *args);
+  return hipGetLastError();
+}
+}

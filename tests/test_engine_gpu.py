@@ -1,0 +1,259 @@
+"""GPU parity tests: HIP engine vs CPU oracle, bit-exact.
+
+These are the parity tests proper: every compute result of the engine
+(labels, VRF nonce, proofs, verification verdicts) is compared byte-for-byte
+with the oracle on the same seeded inputs, at sizes the oracle finishes in
+seconds, plus the OpenSSL-pinned golden labels at mainnet N=8192.
+All calls cross the C-ABI (include/spacemesh_post.h).
+"""
+import ctypes
+import json
+import os
+
+import pytest
+
+import gsm_amd
+from oracle import Oracle, Proof, make_meta
+
+pytestmark = pytest.mark.gpu
+
+NODE = bytes([0xA5]) * 32
+ATX = bytes([0x5A]) * 32
+CHALLENGE = bytes(32)
+POW_DIFF = bytes([0x0F]) + bytes([0xFF]) * 31
+GOLDEN = json.load(open(os.path.join(os.path.dirname(__file__), "golden",
+                                     "golden.json")))
+
+
+def make_mgr(num_units, lpu, scrypt_n, **kw):
+    cfg = gsm_amd.PostConfig(min_num_units=1, labels_per_unit=lpu,
+                             k1=12, k2=8, k3=4, pow_difficulty=POW_DIFF)
+    kw.setdefault("scratch_bytes", 32 << 30)  # keep test allocations modest
+    opts = gsm_amd.PostSetupOpts(num_units=num_units, scrypt_n=scrypt_n,
+                                 **kw)
+    mgr = gsm_amd.PostSetupManager(NODE, ATX, cfg, opts)
+    return cfg, mgr
+
+
+@pytest.fixture(scope="module")
+def small_init():
+    """2^14 labels at N=128 in sink mode, labels kept on device."""
+    cfg, mgr = make_mgr(1, 1 << 14, 128)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    yield cfg, mgr
+    mgr.reset()
+
+
+def test_labels_bit_exact_vs_oracle(small_init):
+    _, mgr = small_init
+    total = 1 << 14
+    got = mgr.copy_labels(0, total)
+    o = Oracle()
+    commit = o.commitment(NODE, ATX)
+    want, _ = o.init_range(commit, 0, total, 128)
+    assert got == want  # byte-for-byte postdata content parity
+
+
+def test_vrf_nonce_matches_oracle(small_init):
+    _, mgr = small_init
+    o = Oracle()
+    commit = o.commitment(NODE, ATX)
+    _, best = o.init_range(commit, 0, 1 << 14, 128)
+    got = mgr.vrf_nonce()
+    assert got is not None
+    assert got[0] == best.index
+    assert got[1] == bytes(best.label)
+
+
+def test_progress_counter_complete(small_init):
+    _, mgr = small_init
+    st = mgr.status()
+    assert st["state"] == mgr.COMPLETE
+    assert st["num_labels_written"] == 1 << 14
+
+
+def test_labels_golden_openssl_n8192():
+    """Mainnet scrypt-N: engine labels == OpenSSL-computed golden vectors."""
+    lv = GOLDEN["labels_openssl"]
+    cfg, mgr = make_mgr(1, 1 << 11, 8192)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    got = mgr.copy_labels(0, 1 << 11)
+    for v in lv["labels"]:
+        if v["N"] == 8192 and v["index"] < (1 << 11):
+            lab = got[v["index"] * 16:(v["index"] + 1) * 16]
+            assert lab.hex() == v["full"][:32], v
+    mgr.reset()
+
+
+def test_file_mode_and_resume(tmp_path):
+    """postdata_*.bin file split + resume-after-interrupt parity
+    (activation/post.go:56,267-271)."""
+    d = str(tmp_path)
+    lpu, n = 1 << 12, 128
+    per_file_labels = 1 << 10
+    cfg, mgr = make_mgr(1, lpu, n, data_dir=d,
+                        max_file_size=per_file_labels * 16)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    files = sorted(f for f in os.listdir(d) if f.startswith("postdata_")
+                   and f.endswith(".bin"))
+    assert len(files) == lpu // per_file_labels
+    data = b"".join(open(os.path.join(d, f), "rb").read() for f in files)
+    o = Oracle()
+    commit = o.commitment(NODE, ATX)
+    want, _ = o.init_range(commit, 0, lpu, n)
+    assert data == want
+    assert os.path.exists(os.path.join(d, "postdata_metadata.json"))
+    mgr.reset()
+
+    # truncate the last file mid-way; a fresh session must resume and
+    # reproduce identical bytes
+    last = os.path.join(d, files[-1])
+    with open(last, "r+b") as f:
+        f.truncate(100 * 16)
+    cfg, mgr2 = make_mgr(1, lpu, n, data_dir=d,
+                         max_file_size=per_file_labels * 16)
+    mgr2.prepare_initializer()
+    st = mgr2.status()
+    assert st["num_labels_written"] == lpu - per_file_labels + 100
+    mgr2.start_session()
+    data2 = b"".join(open(os.path.join(d, f), "rb").read() for f in files)
+    assert data2 == want
+    mgr2.reset()
+
+
+@pytest.fixture(scope="module")
+def roundtrip():
+    """init -> prove on GPU; oracle proves the same labels for comparison."""
+    NU, LPU, N = 2, 1 << 10, 32
+    total = NU * LPU
+    cfg, mgr = make_mgr(NU, LPU, N)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    labels = mgr.copy_labels(0, total)
+    proof = gsm_amd.api.prove_buffer(
+        labels, total, NODE, ATX, CHALLENGE,
+        gsm_amd.PostConfig(k1=12, k2=8, k3=4, pow_difficulty=POW_DIFF),
+        gsm_amd.ProveOpts(nonces=16))
+    yield NU, LPU, N, labels, proof, mgr
+    mgr.reset()
+
+
+def test_proof_bit_exact_vs_oracle(roundtrip):
+    NU, LPU, N, labels, proof, _ = roundtrip
+    o = Oracle()
+    op = o.prove(labels, NU * LPU, CHALLENGE, 12, 8, 16, POW_DIFF)
+    assert proof.nonce == op.nonce
+    assert proof.pow == op.pow
+    assert proof.indices == bytes(op.indices[:op.indices_len])
+
+
+def test_gpu_verify_accepts_gpu_proof(roundtrip):
+    NU, LPU, N, labels, proof, _ = roundtrip
+    cfg = gsm_amd.PostConfig(k1=12, k2=8, k3=8, pow_difficulty=POW_DIFF)
+    ver = gsm_amd.PostVerifier(cfg, scrypt_n=N)
+    meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, NU, LPU)
+    ver.verify(proof, meta)  # full K2
+    # subset + selected-index paths
+    cfg3 = gsm_amd.PostConfig(k1=12, k2=8, k3=3, pow_difficulty=POW_DIFF)
+    ver3 = gsm_amd.PostVerifier(cfg3, scrypt_n=N)
+    ver3.verify(proof, meta,
+                gsm_amd.VerifyOpts(subset_seed=b"peer-seed"))
+    ver.verify(proof, meta, gsm_amd.VerifyOpts(selected_index=0))
+
+
+def test_gpu_verify_verdicts_match_oracle(roundtrip):
+    """Accept/reject parity incl. the adversarial corrupt-index case
+    (systest distributed_post_verification_test.go:253-267)."""
+    NU, LPU, N, labels, proof, _ = roundtrip
+    o = Oracle()
+    meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, NU, LPU)
+    ometa = make_meta(NODE, ATX, CHALLENGE, NU, LPU)
+    cfg = gsm_amd.PostConfig(k1=12, k2=8, k3=8, pow_difficulty=POW_DIFF)
+    ver = gsm_amd.PostVerifier(cfg, scrypt_n=N)
+
+    def o_proof(p):
+        op = Proof()
+        op.nonce = p.nonce
+        op.pow = p.pow
+        op.num_indices = 8
+        op.indices_len = len(p.indices)
+        for i, b in enumerate(p.indices):
+            op.indices[i] = b
+        return op
+
+    # corrupt each byte of the indices in turn; engine and oracle must agree
+    agree = 0
+    for pos in range(len(proof.indices)):
+        bad = bytearray(proof.indices)
+        bad[pos] ^= 0x5A
+        bp = gsm_amd.PostProof(proof.nonce, bytes(bad), proof.pow)
+        orc, _ = o.verify(o_proof(bp), ometa, N, 12, 8, 8, None, -1, POW_DIFF)
+        try:
+            ver.verify(bp, meta)
+            erc = 0
+        except gsm_amd.EngineError as e:
+            erc = {gsm_amd.api.Status.INVALID_INDEX: 1,
+                   gsm_amd.api.Status.POW: 2,
+                   gsm_amd.api.Status.INVALID_ARGS: 3}.get(e.code, -1)
+        assert (orc == 0) == (erc == 0), (pos, orc, erc)
+        if orc == erc:
+            agree += 1
+    assert agree >= len(proof.indices) - 1  # allow err-code drift, not verdict
+
+    # bad pow
+    bp = gsm_amd.PostProof(proof.nonce, proof.indices, proof.pow + 1)
+    with pytest.raises(gsm_amd.EngineError) as ei:
+        ver.verify(bp, meta)
+    assert ei.value.code == gsm_amd.api.Status.POW
+
+
+def test_verify_batch_mixed(roundtrip):
+    NU, LPU, N, labels, proof, _ = roundtrip
+    meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, NU, LPU)
+    bad = gsm_amd.PostProof(proof.nonce,
+                            bytes([proof.indices[0] ^ 0x5A]) +
+                            proof.indices[1:], proof.pow)
+    cfg = gsm_amd.PostConfig(k1=12, k2=8, k3=8, pow_difficulty=POW_DIFF)
+    ver = gsm_amd.PostVerifier(cfg, scrypt_n=N)
+    res = ver.verify_batch([proof, bad, proof], [meta, meta, meta])
+    assert res[0][0] == gsm_amd.api.Status.OK
+    assert res[1][0] != gsm_amd.api.Status.OK
+    assert res[2][0] == gsm_amd.api.Status.OK
+
+
+def test_vrf_nonce_verify_gpu(roundtrip):
+    NU, LPU, N, labels, proof, mgr = roundtrip
+    nonce = mgr.vrf_nonce()
+    assert nonce is not None
+    cfg = gsm_amd.PostConfig(k1=12, k2=8, k3=8, pow_difficulty=POW_DIFF)
+    ver = gsm_amd.PostVerifier(cfg, scrypt_n=N)
+    meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, NU, LPU)
+    ver.verify_vrf_nonce(meta, nonce[0])
+
+
+def test_cancel_mid_init():
+    """Cancellation maps to the Stopped state (post.go:297-301)."""
+    import threading
+    cfg, mgr = make_mgr(1, 1 << 22, 8192, scratch_bytes=8 << 30)
+    mgr.prepare_initializer()
+    t = threading.Timer(0.3, mgr.stop)
+    t.start()
+    with pytest.raises(gsm_amd.EngineError) as ei:
+        mgr.start_session()
+    assert ei.value.code == gsm_amd.api.Status.CANCELLED
+    assert mgr.status()["state"] == mgr.STOPPED
+    t.cancel()
+    mgr.reset()
+
+
+def test_providers_and_benchmark():
+    eng = gsm_amd.Engine()
+    provs = eng.providers()
+    assert len(provs) >= 1
+    assert provs[0]["memory_bytes"] > 0
+    lps = eng.benchmark(0, 8192)
+    assert lps > 0
+    print("benchmark labels/s:", lps)
