@@ -50,7 +50,7 @@ def cpu_baseline():
         if not os.path.exists(bench_bin):
             subprocess.run(["make", "-C", oracle_dir], check=True,
                            capture_output=True, timeout=300)
-        sample = max(1024, 192 * cores)  # ~10-30 s of CPU work
+        sample = max(2048, 1024 * cores)  # ~5-15 s wall on all cores
         out = subprocess.run(
             [bench_bin, "bench", "--labels", str(sample), "--scrypt-n",
              str(SCRYPT_N)],
