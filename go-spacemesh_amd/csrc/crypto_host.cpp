@@ -443,11 +443,13 @@ uint64_t proving_difficulty(uint32_t k1, uint64_t num_labels) {
 }
 
 void vrf_difficulty(uint64_t num_labels, uint8_t out[32]) {
-  if (num_labels <= 1) {
+  /* floor(POSTE_VRF_MARGIN * 2^256 / num_labels): x16 margin so one full
+   * init pass finds a qualifying nonce w.p. 1 - e^-16 (RESTATED) */
+  if (num_labels <= POSTE_VRF_MARGIN) {
     std::memset(out, 0xff, 32);
     return;
   }
-  unsigned __int128 rem = 1;
+  unsigned __int128 rem = POSTE_VRF_MARGIN;
   for (int limb = 0; limb < 4; limb++) {
     rem <<= 64;
     uint64_t q = (uint64_t)(rem / num_labels);

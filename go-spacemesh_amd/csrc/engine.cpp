@@ -18,6 +18,7 @@
 #include <array>
 #include <atomic>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <map>
 #include <mutex>
@@ -73,6 +74,19 @@ void difficulty_to_be_words(const uint8_t d[32], uint32_t w[8]) {
 
 constexpr uint32_t THREADS = 256;
 constexpr uint32_t CAND_CAP = 1 << 16;
+
+/* lookup-gap (kernel_args.h): default gap=2 for real scrypt sizes — halves
+ * the HBM scratch (2x in-flight labels) and write traffic for ~1 extra
+ * BlockMix per phase-2 read at wave level.  POST_GAP_SHIFT overrides. */
+uint32_t pick_gap_shift(uint32_t scrypt_n) {
+  const char *env = getenv("POST_GAP_SHIFT");
+  if (env) {
+    uint32_t g = (uint32_t)atoi(env);
+    while ((scrypt_n >> g) == 0) g--;
+    return g;
+  }
+  return scrypt_n >= 512 ? 1 : 0;
+}
 
 struct DeviceTables { /* AES tables resident per device */
   uint32_t *d_te = nullptr;
@@ -147,6 +161,7 @@ struct PostInitSession {
   uint8_t nonce_label[32];
 
   uint64_t lanes = 0;       /* concurrent labels (scratch slots) */
+  uint32_t gap_shift = 0;
   uint64_t batch = 0;       /* labels per launch */
   uint32_t *d_scratch = nullptr;
   uint8_t *d_out = nullptr; /* batch output (16 B per label) */
@@ -225,19 +240,20 @@ int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
   }
   poste::commitment(cfg->node_id, cfg->commitment_atx_id, s->commitment);
 
-  /* scratch sizing: lanes * 128 * N bytes */
+  /* scratch sizing: lanes * 128 * (N >> gap_shift) bytes */
+  s->gap_shift = pick_gap_shift(cfg->scrypt_n);
   size_t free_b = 0, total_b = 0;
   HIP_TRY(hipMemGetInfo(&free_b, &total_b));
   uint64_t budget = cfg->scratch_bytes
                         ? cfg->scratch_bytes
                         : (uint64_t)((double)free_b * 0.80);
-  uint64_t per_lane = (uint64_t)cfg->scrypt_n * 128;
+  uint64_t per_lane = ((uint64_t)cfg->scrypt_n >> s->gap_shift) * 128;
   uint64_t lanes = budget / per_lane;
   uint64_t range = s->range_end - s->range_start;
   lanes = std::min<uint64_t>(lanes, std::max<uint64_t>(range, THREADS));
-  /* cap lane count: beyond ~8 waves/CU the scratch grows without latency
-   * benefit (256 CU * 2048 threads = 512Ki lanes max) */
-  lanes = std::min<uint64_t>(lanes, 512ull * 1024);
+  /* scratch beyond the kernel's resident-lane capacity is wasted: extra
+   * workgroups only queue (grid-stride covers the batch regardless) */
+  lanes = std::min<uint64_t>(lanes, poste_label_kernel_resident_lanes());
   lanes = (lanes / THREADS) * THREADS;
   if (lanes == 0) {
     delete s;
@@ -375,6 +391,7 @@ int post_init_step(PostInitSession *s, uint64_t max_labels, uint64_t *done) {
   std::memset(&args, 0, sizeof(args));
   load_commitment_words(s->commitment, args.commitment_le);
   args.scrypt_n = s->cfg.scrypt_n;
+  args.gap_shift = s->gap_shift;
   args.out_full = 0;
   args.scratch = s->d_scratch;
   args.scratch_lanes = s->lanes;
@@ -487,6 +504,66 @@ double post_init_last_kernel_ms(const PostInitSession *s) {
   return s ? s->last_kernel_ms : 0.0;
 }
 
+/* nonce-only batch beyond the configured range (no label output) */
+static int nonce_only_batch(PostInitSession *s, uint64_t gstart,
+                            uint64_t count) {
+  LabelKernelArgs args;
+  std::memset(&args, 0, sizeof(args));
+  load_commitment_words(s->commitment, args.commitment_le);
+  args.scrypt_n = s->cfg.scrypt_n;
+  args.gap_shift = s->gap_shift;
+  args.scratch = s->d_scratch;
+  args.scratch_lanes = s->lanes;
+  args.start = gstart;
+  args.count = count;
+  args.out = nullptr;
+  args.has_difficulty = 1;
+  args.cand = s->d_cand;
+  args.cand_count = s->d_cand_count;
+  args.cand_cap = CAND_CAP;
+  uint8_t cur[32];
+  std::memset(cur, 0xff, 32);
+  {
+    std::lock_guard<std::mutex> lk(s->nonce_mu);
+    if (s->nonce_found) std::memcpy(cur, s->nonce_label, 32);
+  }
+  difficulty_to_be_words(cur, args.difficulty_be);
+  unsigned int zero = 0;
+  HIP_TRY(hipMemcpy(s->d_cand_count, &zero, sizeof(zero),
+                    hipMemcpyHostToDevice));
+  HIP_TRY(poste_launch_label_kernel(&args, (uint32_t)(s->lanes / THREADS),
+                                    s->stream));
+  HIP_TRY(hipStreamSynchronize(s->stream));
+  unsigned int n_cand = 0;
+  HIP_TRY(hipMemcpy(&n_cand, s->d_cand_count, sizeof(n_cand),
+                    hipMemcpyDeviceToHost));
+  if (n_cand > 0) {
+    unsigned int take = std::min(n_cand, CAND_CAP);
+    std::vector<PostVrfCandidate> cands(take);
+    HIP_TRY(hipMemcpy(cands.data(), s->d_cand,
+                      sizeof(PostVrfCandidate) * take,
+                      hipMemcpyDeviceToHost));
+    std::lock_guard<std::mutex> lk(s->nonce_mu);
+    for (unsigned int i = 0; i < take; i++) {
+      uint8_t lab[32];
+      for (int k = 0; k < 8; k++) {
+        uint32_t w = cands[i].label_be[k];
+        lab[4 * k] = (uint8_t)(w >> 24);
+        lab[4 * k + 1] = (uint8_t)(w >> 16);
+        lab[4 * k + 2] = (uint8_t)(w >> 8);
+        lab[4 * k + 3] = (uint8_t)w;
+      }
+      int c = s->nonce_found ? std::memcmp(lab, s->nonce_label, 32) : -1;
+      if (c < 0 || (c == 0 && cands[i].index < s->nonce_idx)) {
+        s->nonce_found = true;
+        s->nonce_idx = cands[i].index;
+        std::memcpy(s->nonce_label, lab, 32);
+      }
+    }
+  }
+  return POST_OK;
+}
+
 int post_init_run(PostInitSession *s) {
   if (!s) return POST_ERR_INVALID_ARGS;
   uint64_t range = s->range_end - s->range_start;
@@ -496,6 +573,29 @@ int post_init_run(PostInitSession *s) {
     if (rc != POST_OK) return rc;
     if (done == 0) break;
   }
+
+  /* When this session covers the whole label space and the tracked minimum
+   * does not meet the VRF threshold (p ~ e^-16 with the x16 margin), keep
+   * searching past the end — the reference initializer's follow-up nonce
+   * search (init completes with Nonce optional; VerifyVRFNonce needs one
+   * below threshold, validation.go:261-286). */
+  uint64_t total = (uint64_t)s->cfg.num_units * s->cfg.labels_per_unit;
+  if (s->range_start == 0 && s->range_end == total) {
+    uint8_t diff[32];
+    poste::vrf_difficulty(total, diff);
+    for (uint64_t ext = total; ext < 2 * total; ext += s->batch) {
+      {
+        std::lock_guard<std::mutex> lk(s->nonce_mu);
+        if (s->nonce_found &&
+            std::memcmp(s->nonce_label, diff, 32) < 0)
+          break;
+      }
+      if (s->cancel.load()) return POST_ERR_CANCELLED;
+      int rc = nonce_only_batch(s, ext, std::min(s->batch, 2 * total - ext));
+      if (rc != POST_OK) return rc;
+    }
+  }
+
   if (!s->data_dir.empty()) return write_metadata(s);
   return POST_OK;
 }
@@ -800,9 +900,11 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
   /* scratch-bounded chunks of label recomputes on the GPU */
   size_t free_b = 0, total_b = 0;
   HIP_TRY(hipMemGetInfo(&free_b, &total_b));
-  uint64_t per_lane = (uint64_t)cfg->scrypt_n * 128;
+  uint32_t gap_shift = pick_gap_shift(cfg->scrypt_n);
+  uint64_t per_lane = ((uint64_t)cfg->scrypt_n >> gap_shift) * 128;
   uint64_t max_lanes = (uint64_t)((double)free_b * 0.75) / per_lane;
-  max_lanes = std::min<uint64_t>(max_lanes, 512ull * 1024);
+  max_lanes = std::min<uint64_t>(max_lanes,
+                                 poste_label_kernel_resident_lanes());
   max_lanes = (max_lanes / THREADS) * THREADS;
   if (max_lanes == 0) {
     set_error("not enough memory for verification scratch");
@@ -837,6 +939,7 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
   LabelKernelArgs la;
   std::memset(&la, 0, sizeof(la));
   la.scrypt_n = cfg->scrypt_n;
+  la.gap_shift = gap_shift;
   la.out_full = 1;
   la.scratch = d_scratch;
   la.scratch_lanes = lanes;
@@ -906,7 +1009,8 @@ int post_verify_vrf_nonce(const PostProofMetadata *meta, uint64_t index,
   uint8_t cm[32];
   poste::commitment(meta->node_id, meta->commitment_atx_id, cm);
 
-  uint64_t per_lane = (uint64_t)scrypt_n * 128;
+  uint32_t gap_shift = pick_gap_shift(scrypt_n);
+  uint64_t per_lane = ((uint64_t)scrypt_n >> gap_shift) * 128;
   uint32_t *d_scratch = nullptr;
   uint64_t *d_idx = nullptr;
   uint8_t *d_out = nullptr;
@@ -918,6 +1022,7 @@ int post_verify_vrf_nonce(const PostProofMetadata *meta, uint64_t index,
   std::memset(&la, 0, sizeof(la));
   load_commitment_words(cm, la.commitment_le);
   la.scrypt_n = scrypt_n;
+  la.gap_shift = gap_shift;
   la.out_full = 1;
   la.scratch = d_scratch;
   la.scratch_lanes = THREADS;

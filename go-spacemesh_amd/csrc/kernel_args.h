@@ -12,6 +12,10 @@ struct LabelKernelArgs {
   uint64_t start;
   uint64_t count;
   uint32_t scrypt_n;
+  uint32_t gap_shift; /* lookup-gap = 1<<gap_shift: store every gap-th ROMix
+                         block, recompute the rest on read (SURVEY §7 step 3
+                         HBM-traffic/VALU trade; scratch = N>>gap_shift
+                         blocks per lane) */
   uint32_t out_full;
   uint32_t *scratch;
   uint64_t scratch_lanes;
@@ -45,6 +49,7 @@ hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
                                      uint32_t blocks, hipStream_t stream);
 hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
                                     uint32_t blocks, hipStream_t stream);
+uint64_t poste_label_kernel_resident_lanes(void);
 }
 
 #endif

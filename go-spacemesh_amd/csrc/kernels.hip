@@ -213,35 +213,46 @@ post_label_kernel(LabelKernelArgs a) {
         X[8 * (blk - 1) + k] = __builtin_bswap32(d[k]);
     }
 
-    /* ROMix phase 1: V_j = X; X = BlockMix(X).  Scratch layout: the 128-B
-     * block of iteration j for this lane lives at uint4 offset
-     * (j*lanes + lane)*8 — per-lane bursts stay 128-B contiguous. */
+    /* ROMix phase 1: V_j = X for j % gap == 0; X = BlockMix(X).
+     * Scratch layout: stored block j/gap for this lane lives at uint4
+     * offset ((j/gap)*lanes + lane)*8 — per-lane bursts stay 128-B
+     * contiguous; write traffic = N/gap blocks per label. */
+    const uint32_t gmask = (1u << a.gap_shift) - 1u;
     {
       unsigned long long base = lane * 8ull;
       const unsigned long long stride = a.scratch_lanes * 8ull;
       for (uint32_t j = 0; j < n; j++) {
-        uint4 *p = V + base;
+        if ((j & gmask) == 0) {
+          uint4 *p = V + base;
 #pragma unroll
-        for (int c = 0; c < 8; c++)
-          p[c] = make_uint4(X[4 * c], X[4 * c + 1], X[4 * c + 2],
-                            X[4 * c + 3]);
+          for (int c = 0; c < 8; c++)
+            p[c] = make_uint4(X[4 * c], X[4 * c + 1], X[4 * c + 2],
+                              X[4 * c + 3]);
+          base += stride;
+        }
         blockmix_r1(X);
-        base += stride;
       }
     }
-    /* phase 2: j = Integerify(X) & (n-1); X ^= V_j; BlockMix */
+    /* phase 2: j = Integerify(X) & (n-1); regenerate V_j from the stored
+     * block j & ~gmask by j%gap BlockMixes; X ^= V_j; BlockMix */
     for (uint32_t i = 0; i < n; i++) {
       uint32_t j = X[16] & mask; /* low word suffices: n <= 2^32 */
-      const uint4 *p = V + ((unsigned long long)j * a.scratch_lanes + lane) *
-                               8ull;
+      const uint32_t r = j & gmask;
+      const uint4 *p =
+          V + ((unsigned long long)(j >> a.gap_shift) * a.scratch_lanes +
+               lane) * 8ull;
+      uint32_t Y[32];
 #pragma unroll
       for (int c = 0; c < 8; c++) {
         uint4 v = p[c];
-        X[4 * c] ^= v.x;
-        X[4 * c + 1] ^= v.y;
-        X[4 * c + 2] ^= v.z;
-        X[4 * c + 3] ^= v.w;
+        Y[4 * c] = v.x;
+        Y[4 * c + 1] = v.y;
+        Y[4 * c + 2] = v.z;
+        Y[4 * c + 3] = v.w;
       }
+      for (uint32_t t = 0; t < r; t++) blockmix_r1(Y); /* divergent trip */
+#pragma unroll
+      for (int k = 0; k < 32; k++) X[k] ^= Y[k];
       blockmix_r1(X);
     }
 
@@ -451,6 +462,24 @@ hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
   hipLaunchKernelGGL(post_label_kernel, dim3(blocks), dim3(POSTE_THREADS), 0,
                      stream, *args);
   return hipGetLastError();
+}
+
+/* Max simultaneously-resident lanes of the label kernel on the current
+ * device (occupancy query x CUs x block size).  Scratch beyond this is
+ * wasted memory: extra workgroups just queue. */
+uint64_t poste_label_kernel_resident_lanes(void) {
+  int blocks_per_cu = 0;
+  if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
+          &blocks_per_cu, reinterpret_cast<const void *>(post_label_kernel),
+          POSTE_THREADS, 0) != hipSuccess ||
+      blocks_per_cu <= 0)
+    blocks_per_cu = 2;
+  hipDeviceProp_t prop;
+  int dev = 0;
+  (void)hipGetDevice(&dev);
+  if (hipGetDeviceProperties(&prop, dev) != hipSuccess)
+    return (uint64_t)blocks_per_cu * 256 * POSTE_THREADS;
+  return (uint64_t)blocks_per_cu * prop.multiProcessorCount * POSTE_THREADS;
 }
 
 hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,

@@ -15,6 +15,7 @@
 #define POSTE_NONCES_PER_AES 2 /* 2 nonces per AES cipher */
 #define POSTE_NONCE_GROUP 16   /* k2pow granularity in nonces */
 #define POSTE_K2POW_PREFIX "k2pow" /* 5 bytes, blake3-mode k2pow domain tag */
+#define POSTE_VRF_MARGIN 16        /* vrf threshold = 16*2^256/num_labels */
 
 /* scrypt fixed params on this path (activation/post.go:155): r=1, p=1. */
 #define POSTE_SCRYPT_R 1

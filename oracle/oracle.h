@@ -79,6 +79,7 @@ void oracle_aes128_enc_block(const uint8_t key[16], const uint8_t in[16],
 #define ORACLE_FULL_LABEL_SIZE 32   /* bytes compared for the VRF nonce */
 #define ORACLE_NONCES_PER_AES 2     /* RESTATED: 2 nonces per AES cipher (144 encs @288 nonces) */
 #define ORACLE_NONCE_GROUP 16       /* RESTATED: k2pow granularity in nonces */
+#define ORACLE_VRF_MARGIN 16        /* RESTATED: vrf threshold = 16*2^256/num_labels */
 
 /* commitment = blake3(node_id[32] || commitment_atx_id[32])  (RESTATED) */
 void oracle_commitment(const uint8_t node_id[32],
@@ -110,7 +111,7 @@ int oracle_init_range(const uint8_t commitment[32],
                       const uint8_t difficulty[32],
                       OracleVrfNonce *best);
 
-/* difficulty = floor(2^256 / num_labels) as 32 big-endian bytes (RESTATED) */
+/* difficulty = floor(ORACLE_VRF_MARGIN * 2^256 / num_labels), 32 BE bytes (RESTATED) */
 void oracle_vrf_difficulty(uint64_t num_labels, uint8_t out[32]);
 
 /* proving difficulty = floor(k1 * 2^64 / num_labels)  (RESTATED) */

@@ -74,13 +74,15 @@ int oracle_init_range(const uint8_t commitment[32], uint64_t start,
 }
 
 void oracle_vrf_difficulty(uint64_t num_labels, uint8_t out[32]) {
-  /* floor(2^256 / num_labels), big-endian.  Long division of the 33-byte
-   * value 2^256 by num_labels over 64-bit limbs. (RESTATED) */
-  if (num_labels <= 1) {
+  /* floor(16 * 2^256 / num_labels), big-endian: the nonce threshold with a
+   * x16 margin so a full init pass finds a qualifying label w.p.
+   * 1 - e^-16 (RESTATED; margin constant ORACLE_VRF_MARGIN).  Long division
+   * of 16*2^256 by num_labels over 64-bit limbs. */
+  if (num_labels <= 16) {
     memset(out, 0xff, 32);
     return;
   }
-  unsigned __int128 rem = 1; /* leading limb of 2^256 = [1,0,0,0,0] */
+  unsigned __int128 rem = 16; /* leading limb of 16 * 2^256 */
   for (int limb = 0; limb < 4; limb++) {
     rem <<= 64;
     uint64_t q = (uint64_t)(rem / num_labels);

@@ -168,9 +168,11 @@ def test_proving_difficulty_formula(oracle):
 def test_vrf_difficulty_formula(oracle):
     import ctypes
     out = ctypes.create_string_buffer(32)
-    for n in [2, 128, 1 << 36]:
+    for n in [128, 1 << 22, 1 << 36]:
         oracle.lib.oracle_vrf_difficulty(n, out)
-        assert int.from_bytes(out.raw, "big") == (1 << 256) // n
+        assert int.from_bytes(out.raw, "big") == (16 << 256) // n
+    oracle.lib.oracle_vrf_difficulty(16, out)
+    assert out.raw == b"\xff" * 32
 
 
 def test_k2pow_verify_matches_search(oracle):
