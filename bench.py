@@ -42,6 +42,9 @@ BYTES_PER_LABEL = 128 * SCRYPT_N * 2 + 16
 def cpu_baseline():
     """Time the CPU oracle (the reference CPU provider's role) on this
     box's host cores over a bounded sample."""
+    if os.environ.get("POST_SKIP_CPU_BASELINE"):
+        return {"value": None, "unit": "labels/s", "cores": 0,
+                "kind": "port", "sample": "skipped (A/B run)"}
     try:
         import multiprocessing
         cores = multiprocessing.cpu_count()
