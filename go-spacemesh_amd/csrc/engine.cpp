@@ -160,7 +160,8 @@ struct PostInitSession {
   uint64_t nonce_idx = 0;
   uint8_t nonce_label[32];
 
-  uint64_t lanes = 0;       /* concurrent labels (scratch slots) */
+  uint64_t lanes = 0;       /* concurrent labels (scratch slots); one slot
+                               = one quad of 4 threads in the kernel */
   uint32_t gap_shift = 0;
   uint64_t batch = 0;       /* labels per launch */
   uint32_t *d_scratch = nullptr;
@@ -250,11 +251,12 @@ int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
   uint64_t per_lane = ((uint64_t)cfg->scrypt_n >> s->gap_shift) * 128;
   uint64_t lanes = budget / per_lane;
   uint64_t range = s->range_end - s->range_start;
-  lanes = std::min<uint64_t>(lanes, std::max<uint64_t>(range, THREADS));
-  /* scratch beyond the kernel's resident-lane capacity is wasted: extra
-   * workgroups only queue (grid-stride covers the batch regardless) */
-  lanes = std::min<uint64_t>(lanes, poste_label_kernel_resident_lanes());
-  lanes = (lanes / THREADS) * THREADS;
+  lanes = std::min<uint64_t>(lanes, std::max<uint64_t>(range, 64));
+  /* scratch beyond the kernel's resident capacity is wasted: extra
+   * workgroups only queue (grid-stride covers the batch regardless).
+   * resident threads / 4 = resident quads (labels in flight). */
+  lanes = std::min<uint64_t>(lanes, poste_label_kernel_resident_lanes() / 4);
+  lanes = (lanes / 64) * 64; /* 64 quads = one 256-thread workgroup */
   if (lanes == 0) {
     delete s;
     set_error("not enough device memory for one scratch lane block");
@@ -410,7 +412,7 @@ int post_init_step(PostInitSession *s, uint64_t max_labels, uint64_t *done) {
     if (s->nonce_found) std::memcpy(cur_difficulty, s->nonce_label, 32);
   }
 
-  const uint32_t blocks = (uint32_t)(s->lanes / THREADS);
+  const uint32_t blocks = (uint32_t)(s->lanes / 64); /* 64 quads/block */
   std::vector<PostVrfCandidate> cands(CAND_CAP);
 
   if (!s->ev0) {
@@ -531,7 +533,7 @@ static int nonce_only_batch(PostInitSession *s, uint64_t gstart,
   unsigned int zero = 0;
   HIP_TRY(hipMemcpy(s->d_cand_count, &zero, sizeof(zero),
                     hipMemcpyHostToDevice));
-  HIP_TRY(poste_launch_label_kernel(&args, (uint32_t)(s->lanes / THREADS),
+  HIP_TRY(poste_launch_label_kernel(&args, (uint32_t)(s->lanes / 64),
                                     s->stream));
   HIP_TRY(hipStreamSynchronize(s->stream));
   unsigned int n_cand = 0;
@@ -904,8 +906,8 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
   uint64_t per_lane = ((uint64_t)cfg->scrypt_n >> gap_shift) * 128;
   uint64_t max_lanes = (uint64_t)((double)free_b * 0.75) / per_lane;
   max_lanes = std::min<uint64_t>(max_lanes,
-                                 poste_label_kernel_resident_lanes());
-  max_lanes = (max_lanes / THREADS) * THREADS;
+                                 poste_label_kernel_resident_lanes() / 4);
+  max_lanes = (max_lanes / 64) * 64;
   if (max_lanes == 0) {
     set_error("not enough memory for verification scratch");
     return POST_ERR_OOM;
@@ -917,9 +919,8 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
     h_idx[i] = tasks[i].label_index;
     h_cid[i] = tasks[i].proof;
   }
-  uint64_t lanes = std::min<uint64_t>(max_lanes,
-                                      ((tasks.size() + THREADS - 1) / THREADS) *
-                                          THREADS);
+  uint64_t lanes = std::min<uint64_t>(
+      max_lanes, ((tasks.size() + 63) / 64) * 64);
   uint32_t *d_scratch = nullptr;
   uint64_t *d_idx = nullptr;
   uint32_t *d_cid = nullptr, *d_cm = nullptr;
@@ -948,8 +949,7 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
   la.commit_ids = d_cid;
   la.commitments = d_cm;
   la.count = tasks.size();
-  HIP_TRY(poste_launch_label_kernel(&la, (uint32_t)(lanes / THREADS),
-                                    nullptr));
+  HIP_TRY(poste_launch_label_kernel(&la, (uint32_t)(lanes / 64), nullptr));
   HIP_TRY(hipDeviceSynchronize());
   std::vector<uint8_t> full((size_t)tasks.size() * 32);
   HIP_TRY(hipMemcpy(full.data(), d_out, full.size(),
@@ -1014,7 +1014,7 @@ int post_verify_vrf_nonce(const PostProofMetadata *meta, uint64_t index,
   uint32_t *d_scratch = nullptr;
   uint64_t *d_idx = nullptr;
   uint8_t *d_out = nullptr;
-  HIP_TRY(hipMalloc(&d_scratch, (size_t)THREADS * per_lane));
+  HIP_TRY(hipMalloc(&d_scratch, (size_t)64 * per_lane));
   HIP_TRY(hipMalloc(&d_idx, 8));
   HIP_TRY(hipMalloc(&d_out, 32));
   HIP_TRY(hipMemcpy(d_idx, &index, 8, hipMemcpyHostToDevice));
@@ -1025,7 +1025,7 @@ int post_verify_vrf_nonce(const PostProofMetadata *meta, uint64_t index,
   la.gap_shift = gap_shift;
   la.out_full = 1;
   la.scratch = d_scratch;
-  la.scratch_lanes = THREADS;
+  la.scratch_lanes = 64;
   la.out = d_out;
   la.indices = d_idx;
   la.count = 1;
