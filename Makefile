@@ -10,12 +10,19 @@ CSRC    := go-spacemesh_amd/csrc
 
 all: $(ENGINE) oracle
 
-$(ENGINE): $(CSRC)/kernels.hip $(CSRC)/engine.cpp $(CSRC)/crypto_host.cpp \
-           $(CSRC)/post_common.h $(CSRC)/kernel_args.h $(CSRC)/crypto_host.h \
-           include/spacemesh_post.h
-	$(HIPCC) $(HIPFLAGS) -shared \
-	    $(CSRC)/kernels.hip $(CSRC)/engine.cpp $(CSRC)/crypto_host.cpp \
-	    -o $@
+SRCS_ENGINE := $(CSRC)/kernels.hip $(CSRC)/engine.cpp $(CSRC)/crypto_host.cpp
+HDRS_ENGINE := $(CSRC)/post_common.h $(CSRC)/kernel_args.h \
+               $(CSRC)/crypto_host.h include/spacemesh_post.h
+
+$(ENGINE): $(SRCS_ENGINE) $(HDRS_ENGINE)
+	$(HIPCC) $(HIPFLAGS) -shared $(SRCS_ENGINE) -o $@
+
+# A/B variant: label kernel constrained to >=4 waves/SIMD
+go-spacemesh_amd/libpost_hip_lb4.so: $(SRCS_ENGINE) $(HDRS_ENGINE)
+	$(HIPCC) $(HIPFLAGS) -DPOSTE_LABEL_MIN_WAVES=4 -shared $(SRCS_ENGINE) -o $@
+
+ab: $(ENGINE) go-spacemesh_amd/libpost_hip_lb4.so
+.PHONY: ab
 
 oracle:
 	$(MAKE) -C oracle

@@ -23,7 +23,8 @@ from ctypes import (POINTER, byref, c_char_p, c_int, c_int32, c_size_t,
 from typing import Optional
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
-_LIB_PATH = os.path.join(_DIR, "libpost_hip.so")
+_LIB_PATH = os.environ.get("POST_ENGINE_LIB",
+                           os.path.join(_DIR, "libpost_hip.so"))
 
 LABEL_SIZE = 16
 FULL_LABEL_SIZE = 32

@@ -28,6 +28,13 @@
 
 #define POSTE_THREADS 256
 
+/* minimum waves/SIMD forced on the label kernel (register budget): 2 -> 143
+ * VGPRs no spills; 4 -> 128 VGPRs with spills confined to the per-label SHA
+ * prologue/tail.  Default 2; overridden with -DPOSTE_LABEL_MIN_WAVES=4. */
+#ifndef POSTE_LABEL_MIN_WAVES
+#define POSTE_LABEL_MIN_WAVES 2
+#endif
+
 /* ------------------------- SHA-256 (device) ------------------------- */
 __constant__ uint32_t c_sha_k[64] = {
     0x428a2f98, 0x71374491, 0xb5c0fbcf, 0xe9b5dba5, 0x3956c25b, 0x59f111f1,
@@ -91,11 +98,14 @@ __device__ __forceinline__ void sha_init(uint32_t h[8]) {
  * (measured 7.7 TB/s random vs 1.2 TB/s for the 1-lane 16-B pattern —
  * profiles/probe2 gather1 vs gather4). */
 
-#define SWZ(v, pat) __builtin_amdgcn_ds_swizzle((v), (pat))
-#define QROT1 0x8039 /* quad perm (1,2,3,0): lane l reads elem (l+1)&3 */
-#define QROT2 0x804E /* (2,3,0,1) */
-#define QROT3 0x8093 /* (3,0,1,2) */
-#define QBCAST0 0x8000 /* all lanes read elem 0 */
+/* quad permutes via DPP (VALU pipe, ~2-cycle) instead of ds_swizzle
+ * (LDS pipe, ~50-cycle latency in the salsa dependency chain) */
+#define SWZ(v, pat)                                                            \
+  ((uint32_t)__builtin_amdgcn_mov_dpp((int)(v), (pat), 0xf, 0xf, true))
+#define QROT1 0x39 /* quad perm (1,2,3,0): lane l reads elem (l+1)&3 */
+#define QROT2 0x4E /* (2,3,0,1) */
+#define QROT3 0x93 /* (3,0,1,2) */
+#define QBCAST0 0x00 /* all lanes read elem 0 */
 
 /* z-vector salsa20/8 on one quad: A..D are this lane's elements of z0..z3 */
 __device__ __forceinline__ void salsa8_z(uint32_t &A, uint32_t &B,
@@ -146,22 +156,17 @@ __device__ __forceinline__ void canon_to_z(const uint32_t x[16], uint32_t sub,
   z[0] = s0; z[1] = s1; z[2] = s2; z[3] = s3;
 }
 
-/* this quad's z vectors -> canonical 16 words (every lane gets all 16) */
-__device__ __forceinline__ void z_to_canon(const uint32_t z[4],
-                                           uint32_t x[16]) {
-  uint32_t e[4][4];
+/* this quad's z vectors -> big-endian SHA message words of the canonical
+ * 64-B block (fused broadcast+byteswap, no staging buffers) */
+__device__ __forceinline__ void z_to_msg_be(const uint32_t z[4],
+                                            uint32_t m[16]) {
 #pragma unroll
   for (int v = 0; v < 4; v++) {
-    e[v][0] = SWZ(z[v], 0x8000);
-    e[v][1] = SWZ(z[v], 0x8055);
-    e[v][2] = SWZ(z[v], 0x80AA);
-    e[v][3] = SWZ(z[v], 0x80FF);
+    m[(4 * v + 0) & 15] = __builtin_bswap32(SWZ(z[v], 0x00));
+    m[(4 * v + 5) & 15] = __builtin_bswap32(SWZ(z[v], 0x55));
+    m[(4 * v + 10) & 15] = __builtin_bswap32(SWZ(z[v], 0xAA));
+    m[(4 * v + 15) & 15] = __builtin_bswap32(SWZ(z[v], 0xFF));
   }
-  /* x[(4v + 5l) % 16] = z_v[l] */
-#pragma unroll
-  for (int v = 0; v < 4; v++)
-#pragma unroll
-    for (int l = 0; l < 4; l++) x[(4 * v + 5 * l) & 15] = e[v][l];
 }
 
 /* ------------------------- label kernel ------------------------- */
@@ -185,7 +190,7 @@ __device__ __forceinline__ void hmac_outer(const uint32_t ho[8],
   for (int i = 0; i < 8; i++) out[i] = h[i];
 }
 
-__global__ void __launch_bounds__(POSTE_THREADS)
+__global__ void __launch_bounds__(POSTE_THREADS, POSTE_LABEL_MIN_WAVES)
 post_label_kernel(LabelKernelArgs a) {
   const unsigned long long lane =
       (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -236,7 +241,6 @@ post_label_kernel(LabelKernelArgs a) {
     /* B = PBKDF2(P, "", 1, 128) -> canonical LE salsa words (every lane of
      * the quad computes the same SHA chain — redundant but register-cheap;
      * the SHA work is ~0.1% of the ROMix cost) */
-    uint32_t Xc[16]; /* canonical words of one 64-B block at a time */
     uint32_t Z0[4], Z1[4]; /* this lane's z elements of B0 and B1 */
 #pragma unroll
     for (uint32_t half = 0; half < 2; half++) {
@@ -308,13 +312,9 @@ post_label_kernel(LabelKernelArgs a) {
     uint32_t h[8];
 #pragma unroll
     for (int i = 0; i < 8; i++) h[i] = hi[i];
-    z_to_canon(Z0, Xc);
-#pragma unroll
-    for (int i = 0; i < 16; i++) m[i] = __builtin_bswap32(Xc[i]);
+    z_to_msg_be(Z0, m);
     sha_compress(h, m);
-    z_to_canon(Z1, Xc);
-#pragma unroll
-    for (int i = 0; i < 16; i++) m[i] = __builtin_bswap32(Xc[i]);
+    z_to_msg_be(Z1, m);
     sha_compress(h, m);
     m[0] = 1;
     m[1] = 0x80000000u;
