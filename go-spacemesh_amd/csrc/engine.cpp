@@ -170,6 +170,7 @@ struct PostInitSession {
   bool keep_all = false;
   PostVrfCandidate *d_cand = nullptr;
   unsigned int *d_cand_count = nullptr;
+  uint32_t *d_xbuf = nullptr; /* 128 B per batch task (3-kernel staging) */
   uint8_t *h_batch = nullptr; /* pinned */
   hipStream_t stream = nullptr;
   hipEvent_t ev0 = nullptr, ev1 = nullptr;
@@ -184,6 +185,7 @@ struct PostInitSession {
     if (d_all) (void)hipFree(d_all);
     if (d_cand) (void)hipFree(d_cand);
     if (d_cand_count) (void)hipFree(d_cand_count);
+    if (d_xbuf) (void)hipFree(d_xbuf);
     if (h_batch) (void)hipHostFree(h_batch);
   }
 };
@@ -273,6 +275,7 @@ int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
   }
   HIP_TRY(hipMalloc(&s->d_out, (size_t)s->batch * POST_LABEL_SIZE));
   HIP_TRY(hipMalloc(&s->d_cand, sizeof(PostVrfCandidate) * CAND_CAP));
+  HIP_TRY(hipMalloc(&s->d_xbuf, (size_t)s->batch * 128));
   HIP_TRY(hipMalloc(&s->d_cand_count, sizeof(unsigned int)));
   HIP_TRY(hipHostMalloc(&s->h_batch, (size_t)s->batch * POST_LABEL_SIZE));
   HIP_TRY(hipStreamCreate(&s->stream));
@@ -394,6 +397,7 @@ int post_init_step(PostInitSession *s, uint64_t max_labels, uint64_t *done) {
   load_commitment_words(s->commitment, args.commitment_le);
   args.scrypt_n = s->cfg.scrypt_n;
   args.gap_shift = s->gap_shift;
+  args.xbuf = s->d_xbuf;
   args.out_full = 0;
   args.scratch = s->d_scratch;
   args.scratch_lanes = s->lanes;
@@ -514,6 +518,7 @@ static int nonce_only_batch(PostInitSession *s, uint64_t gstart,
   load_commitment_words(s->commitment, args.commitment_le);
   args.scrypt_n = s->cfg.scrypt_n;
   args.gap_shift = s->gap_shift;
+  args.xbuf = s->d_xbuf;
   args.scratch = s->d_scratch;
   args.scratch_lanes = s->lanes;
   args.start = gstart;
@@ -925,7 +930,9 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
   uint64_t *d_idx = nullptr;
   uint32_t *d_cid = nullptr, *d_cm = nullptr;
   uint8_t *d_out = nullptr;
+  uint32_t *d_xbuf = nullptr;
   HIP_TRY(hipMalloc(&d_scratch, (size_t)lanes * per_lane));
+  HIP_TRY(hipMalloc(&d_xbuf, (size_t)tasks.size() * 128));
   HIP_TRY(hipMalloc(&d_idx, h_idx.size() * 8));
   HIP_TRY(hipMalloc(&d_cid, h_cid.size() * 4));
   HIP_TRY(hipMalloc(&d_cm, commit_words.size() * 4));
@@ -941,6 +948,7 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
   std::memset(&la, 0, sizeof(la));
   la.scrypt_n = cfg->scrypt_n;
   la.gap_shift = gap_shift;
+  la.xbuf = d_xbuf;
   la.out_full = 1;
   la.scratch = d_scratch;
   la.scratch_lanes = lanes;
@@ -955,6 +963,7 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
   HIP_TRY(hipMemcpy(full.data(), d_out, full.size(),
                     hipMemcpyDeviceToHost));
   (void)hipFree(d_scratch);
+  (void)hipFree(d_xbuf);
   (void)hipFree(d_idx);
   (void)hipFree(d_cid);
   (void)hipFree(d_cm);
@@ -1014,15 +1023,18 @@ int post_verify_vrf_nonce(const PostProofMetadata *meta, uint64_t index,
   uint32_t *d_scratch = nullptr;
   uint64_t *d_idx = nullptr;
   uint8_t *d_out = nullptr;
+  uint32_t *d_xbuf1 = nullptr;
   HIP_TRY(hipMalloc(&d_scratch, (size_t)64 * per_lane));
   HIP_TRY(hipMalloc(&d_idx, 8));
   HIP_TRY(hipMalloc(&d_out, 32));
+  HIP_TRY(hipMalloc(&d_xbuf1, 128));
   HIP_TRY(hipMemcpy(d_idx, &index, 8, hipMemcpyHostToDevice));
   LabelKernelArgs la;
   std::memset(&la, 0, sizeof(la));
   load_commitment_words(cm, la.commitment_le);
   la.scrypt_n = scrypt_n;
   la.gap_shift = gap_shift;
+  la.xbuf = d_xbuf1;
   la.out_full = 1;
   la.scratch = d_scratch;
   la.scratch_lanes = 64;
@@ -1036,6 +1048,7 @@ int post_verify_vrf_nonce(const PostProofMetadata *meta, uint64_t index,
   (void)hipFree(d_scratch);
   (void)hipFree(d_idx);
   (void)hipFree(d_out);
+  (void)hipFree(d_xbuf1);
   uint8_t difficulty[32];
   poste::vrf_difficulty(num_labels, difficulty);
   if (std::memcmp(full, difficulty, 32) < 0) return POST_OK;

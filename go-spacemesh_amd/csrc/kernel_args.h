@@ -23,6 +23,9 @@ struct LabelKernelArgs {
   const uint64_t *indices;
   const uint32_t *commit_ids;
   const uint32_t *commitments;
+  uint32_t *xbuf; /* staging for the per-label working block X between the
+                     prologue/romix/tail kernels: 8 uint4 per task, quad-z
+                     chunk layout (task*8 + sub, task*8 + sub + 4) */
   uint32_t has_difficulty;
   uint32_t difficulty_be[8];
   PostVrfCandidate *cand;

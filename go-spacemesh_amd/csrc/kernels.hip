@@ -1,16 +1,13 @@
 /* kernels.hip — MI355X (gfx950/CDNA4) kernels for the POST hot path.
  *
- * post_label_kernel: scrypt-N labeling (the init hot loop the reference
- *   reaches at activation/post.go:295).  One lane = one label: the ROMix
- *   working block X (2x64 B) lives in VGPRs, the N-entry scratchpad V
- *   (128 B * N per in-flight label; 1 MiB at mainnet N=8192) lives in HBM,
- *   block-interleaved across lanes at 128-B granularity so each lane's
- *   per-iteration access is one fully-used 128-B burst (phase-1 stores merge
- *   to full lines in the XCD L2; phase-2 reads are random 128-B gathers —
- *   the HBM-traffic roofline of SURVEY.md §8(d): ~2 MiB/label).
- *   No MFMA: this is a u32 add/xor/rotate hash loop (BASELINE.json
- *   north_star), VALU ~1.36e7 int ops/label.
- *   Also used in index-list mode for verification's label recompute
+ * Labeling (the init hot loop the reference reaches at
+ *   activation/post.go:295): one label per 4-lane quad, run as a 3-kernel
+ *   pipeline (SHA prologue -> register-lean ROMix at 8 waves/SIMD -> SHA
+ *   tail + VRF reduce), working block staged through xbuf.  The N-entry
+ *   ROMix scratchpad V (128 B * N per in-flight label at gap 1) lives in
+ *   HBM; a quad's block accesses are aligned 64-B transactions.  No MFMA:
+ *   a u32 add/xor/rotate hash loop (BASELINE.json north_star).  Also used
+ *   in index-list mode for verification's label recompute
  *   (validation.go:182-222 -> K3 sampled indices) with per-task commitments.
  *
  * post_scan_kernel: the proving index scan (post-service GenProof,
@@ -190,24 +187,46 @@ __device__ __forceinline__ void hmac_outer(const uint32_t ho[8],
   for (int i = 0; i < 8; i++) out[i] = h[i];
 }
 
-__global__ void __launch_bounds__(POSTE_THREADS, POSTE_LABEL_MIN_WAVES)
-post_label_kernel(LabelKernelArgs a) {
+/* The labeling path runs as THREE kernels per batch so the ROMix hot loop
+ * stays register-lean (8 waves/SIMD) while the SHA-heavy prologue/tail keep
+ * their own (lower) occupancy.  The per-label working block X round-trips
+ * through xbuf (128 B per task, quad-z chunk layout) — 256 B of extra
+ * traffic against the ~2 MiB/label scratch stream. */
+
+/* shared helper: password = commitment || LE64(index) -> HMAC states */
+__device__ __forceinline__ void hmac_states_for(const uint32_t *cw,
+                                                unsigned long long index,
+                                                uint32_t hi[8],
+                                                uint32_t ho[8]) {
+  uint32_t pw_be[10], m[16];
+#pragma unroll
+  for (int i = 0; i < 8; i++) pw_be[i] = __builtin_bswap32(cw[i]);
+  pw_be[8] = __builtin_bswap32((uint32_t)index);
+  pw_be[9] = __builtin_bswap32((uint32_t)(index >> 32));
+  sha_init(hi);
+#pragma unroll
+  for (int i = 0; i < 10; i++) m[i] = pw_be[i] ^ 0x36363636u;
+#pragma unroll
+  for (int i = 10; i < 16; i++) m[i] = 0x36363636u;
+  sha_compress(hi, m);
+  sha_init(ho);
+#pragma unroll
+  for (int i = 0; i < 10; i++) m[i] = pw_be[i] ^ 0x5c5c5c5cu;
+#pragma unroll
+  for (int i = 10; i < 16; i++) m[i] = 0x5c5c5c5cu;
+  sha_compress(ho, m);
+}
+
+/* prologue: PBKDF2(P, "", 1, 128) -> xbuf in quad-z layout.  Every lane of
+ * a quad computes the same SHA chain (redundant, ~0.1% of the ROMix cost)
+ * and keeps only its own z slice. */
+__global__ void __launch_bounds__(POSTE_THREADS)
+post_label_prologue_kernel(LabelKernelArgs a) {
   const unsigned long long lane =
       (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
-  const unsigned long long group = lane >> 2; /* quad = one label */
+  const unsigned long long group = lane >> 2;
   const uint32_t sub = (uint32_t)(lane & 3);
-  const uint32_t n = a.scrypt_n;
-  const uint32_t mask = n - 1;
-  uint4 *V = (uint4 *)a.scratch;
-
-  /* per-lane running VRF minimum across this lane's tasks (only sub==0
-   * lanes track); reduced once per workgroup at kernel end so the candidate
-   * buffer stays bounded and the tracked minimum is exact (the
-   * initializer's nonce search, activation/post.go:295, keeps the global
-   * argmin label) */
-  uint32_t min_lab[8];
-  unsigned long long min_idx = 0;
-  int min_found = 0;
+  uint4 *X = (uint4 *)a.xbuf;
 
   for (unsigned long long task = group; task < a.count;
        task += a.scratch_lanes) {
@@ -216,32 +235,9 @@ post_label_kernel(LabelKernelArgs a) {
     const uint32_t *cw = a.commit_ids
                              ? a.commitments + 8ull * a.commit_ids[task]
                              : a.commitment_le;
-    /* password = commitment(32) || LE64(index): 10 LE words -> BE msg words */
-    uint32_t pw_be[10];
-#pragma unroll
-    for (int i = 0; i < 8; i++) pw_be[i] = __builtin_bswap32(cw[i]);
-    pw_be[8] = __builtin_bswap32((uint32_t)index);
-    pw_be[9] = __builtin_bswap32((uint32_t)(index >> 32));
-
-    /* HMAC ipad/opad chaining states for this password */
     uint32_t hi[8], ho[8], m[16];
-    sha_init(hi);
-#pragma unroll
-    for (int i = 0; i < 10; i++) m[i] = pw_be[i] ^ 0x36363636u;
-#pragma unroll
-    for (int i = 10; i < 16; i++) m[i] = 0x36363636u;
-    sha_compress(hi, m);
-    sha_init(ho);
-#pragma unroll
-    for (int i = 0; i < 10; i++) m[i] = pw_be[i] ^ 0x5c5c5c5cu;
-#pragma unroll
-    for (int i = 10; i < 16; i++) m[i] = 0x5c5c5c5cu;
-    sha_compress(ho, m);
-
-    /* B = PBKDF2(P, "", 1, 128) -> canonical LE salsa words (every lane of
-     * the quad computes the same SHA chain — redundant but register-cheap;
-     * the SHA work is ~0.1% of the ROMix cost) */
-    uint32_t Z0[4], Z1[4]; /* this lane's z elements of B0 and B1 */
+    hmac_states_for(cw, index, hi, ho);
+    uint32_t Z0[4], Z1[4];
 #pragma unroll
     for (uint32_t half = 0; half < 2; half++) {
       uint32_t cblk[16];
@@ -265,12 +261,37 @@ post_label_kernel(LabelKernelArgs a) {
       }
       canon_to_z(cblk, sub, half == 0 ? Z0 : Z1);
     }
+    uint4 *p = X + task * 8ull;
+    p[sub] = make_uint4(Z0[0], Z0[1], Z0[2], Z0[3]);
+    p[sub + 4] = make_uint4(Z1[0], Z1[1], Z1[2], Z1[3]);
+  }
+}
 
-    /* ROMix phase 1: V_j = X for j % gap == 0; X = BlockMix(X).
-     * Scratch layout: stored block j/gap of this quad is 8 consecutive
-     * uint4 at ((j/gap)*groups + group)*8; lane sub covers chunks sub and
-     * sub+4, so each wave instruction is aligned 64-B transactions. */
-    const uint32_t gmask = (1u << a.gap_shift) - 1u;
+/* the ROMix hot loop: register-lean, forced to 8 waves/SIMD */
+__global__ void __launch_bounds__(POSTE_THREADS, 8)
+post_label_romix_kernel(LabelKernelArgs a) {
+  const unsigned long long lane =
+      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned long long group = lane >> 2;
+  const uint32_t sub = (uint32_t)(lane & 3);
+  const uint32_t n = a.scrypt_n;
+  const uint32_t mask = n - 1;
+  const uint32_t gmask = (1u << a.gap_shift) - 1u;
+  uint4 *V = (uint4 *)a.scratch;
+  uint4 *X = (uint4 *)a.xbuf;
+
+  for (unsigned long long task = group; task < a.count;
+       task += a.scratch_lanes) {
+    uint32_t Z0[4], Z1[4];
+    {
+      uint4 *p = X + task * 8ull;
+      uint4 v0 = p[sub], v1 = p[sub + 4];
+      Z0[0] = v0.x; Z0[1] = v0.y; Z0[2] = v0.z; Z0[3] = v0.w;
+      Z1[0] = v1.x; Z1[1] = v1.y; Z1[2] = v1.z; Z1[3] = v1.w;
+    }
+    /* phase 1: V_j = X for j % gap == 0; X = BlockMix(X).  Stored block
+     * j/gap of this quad is 8 consecutive uint4 at ((j/gap)*slots+group)*8;
+     * lane sub covers chunks sub and sub+4 -> aligned 64-B transactions. */
     {
       unsigned long long base = group * 8ull + sub;
       const unsigned long long stride = a.scratch_lanes * 8ull;
@@ -285,8 +306,8 @@ post_label_kernel(LabelKernelArgs a) {
       }
     }
     /* phase 2: j = Integerify(X) & (n-1); regenerate V_j from the stored
-     * block j & ~gmask by j%gap BlockMixes; X ^= V_j; BlockMix.
-     * Integerify = canonical word 16 = B1's z0[0] -> quad broadcast. */
+     * block by j%gap BlockMixes; X ^= V_j; BlockMix.  Integerify =
+     * canonical word 16 = B1's z0[0] -> quad broadcast. */
     for (uint32_t i = 0; i < n; i++) {
       uint32_t j = SWZ(Z1[0], QBCAST0) & mask;
       const uint32_t r = j & gmask;
@@ -307,8 +328,44 @@ post_label_kernel(LabelKernelArgs a) {
       }
       blockmix_z(Z0, Z1);
     }
+    {
+      uint4 *p = X + task * 8ull;
+      p[sub] = make_uint4(Z0[0], Z0[1], Z0[2], Z0[3]);
+      p[sub + 4] = make_uint4(Z1[0], Z1[1], Z1[2], Z1[3]);
+    }
+  }
+}
 
-    /* out = PBKDF2(P, X-bytes, 1, 32): reassemble canonical block bytes */
+/* tail: PBKDF2(P, X, 1, 32) -> labels, VRF-minimum tracking + exact
+ * per-workgroup candidate reduce */
+__global__ void __launch_bounds__(POSTE_THREADS)
+post_label_tail_kernel(LabelKernelArgs a) {
+  const unsigned long long lane =
+      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned long long group = lane >> 2;
+  const uint32_t sub = (uint32_t)(lane & 3);
+  uint4 *X = (uint4 *)a.xbuf;
+
+  uint32_t min_lab[8];
+  unsigned long long min_idx = 0;
+  int min_found = 0;
+
+  for (unsigned long long task = group; task < a.count;
+       task += a.scratch_lanes) {
+    const unsigned long long index = a.indices ? a.indices[task]
+                                               : a.start + task;
+    const uint32_t *cw = a.commit_ids
+                             ? a.commitments + 8ull * a.commit_ids[task]
+                             : a.commitment_le;
+    uint32_t hi[8], ho[8], m[16];
+    hmac_states_for(cw, index, hi, ho);
+    uint32_t Z0[4], Z1[4];
+    {
+      uint4 *p = X + task * 8ull;
+      uint4 v0 = p[sub], v1 = p[sub + 4];
+      Z0[0] = v0.x; Z0[1] = v0.y; Z0[2] = v0.z; Z0[3] = v0.w;
+      Z1[0] = v1.x; Z1[1] = v1.y; Z1[2] = v1.z; Z1[3] = v1.w;
+    }
     uint32_t h[8];
 #pragma unroll
     for (int i = 0; i < 8; i++) h[i] = hi[i];
@@ -399,7 +456,7 @@ post_label_kernel(LabelKernelArgs a) {
           continue;
         }
         /* order by (label bytes, index) ascending */
-        int cmp = 0; /* -1: t < best, 0: equal labels, 1: t > best */
+        int cmp = 0;
         for (int k = 0; k < 8 && cmp == 0; k++) {
           if (s_lab[t][k] < s_lab[best][k]) cmp = -1;
           else if (s_lab[t][k] > s_lab[best][k]) cmp = 1;
@@ -509,8 +566,12 @@ extern "C" {
 
 hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
                                      uint32_t blocks, hipStream_t stream) {
-  hipLaunchKernelGGL(post_label_kernel, dim3(blocks), dim3(POSTE_THREADS), 0,
-                     stream, *args);
+  hipLaunchKernelGGL(post_label_prologue_kernel, dim3(blocks),
+                     dim3(POSTE_THREADS), 0, stream, *args);
+  hipLaunchKernelGGL(post_label_romix_kernel, dim3(blocks),
+                     dim3(POSTE_THREADS), 0, stream, *args);
+  hipLaunchKernelGGL(post_label_tail_kernel, dim3(blocks),
+                     dim3(POSTE_THREADS), 0, stream, *args);
   return hipGetLastError();
 }
 
@@ -520,7 +581,8 @@ hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
 uint64_t poste_label_kernel_resident_lanes(void) {
   int blocks_per_cu = 0;
   if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
-          &blocks_per_cu, reinterpret_cast<const void *>(post_label_kernel),
+          &blocks_per_cu,
+          reinterpret_cast<const void *>(post_label_romix_kernel),
           POSTE_THREADS, 0) != hipSuccess ||
       blocks_per_cu <= 0)
     blocks_per_cu = 2;
