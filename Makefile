@@ -17,13 +17,6 @@ HDRS_ENGINE := $(CSRC)/post_common.h $(CSRC)/kernel_args.h \
 $(ENGINE): $(SRCS_ENGINE) $(HDRS_ENGINE)
 	$(HIPCC) $(HIPFLAGS) -shared $(SRCS_ENGINE) -o $@
 
-# A/B variant: label kernel constrained to >=4 waves/SIMD
-go-spacemesh_amd/libpost_hip_lb4.so: $(SRCS_ENGINE) $(HDRS_ENGINE)
-	$(HIPCC) $(HIPFLAGS) -DPOSTE_LABEL_MIN_WAVES=4 -shared $(SRCS_ENGINE) -o $@
-
-ab: $(ENGINE) go-spacemesh_amd/libpost_hip_lb4.so
-.PHONY: ab
-
 oracle:
 	$(MAKE) -C oracle
 

@@ -35,8 +35,9 @@ HBM_PEAK_GBS = 8000.0          # gfx950 spec peak (MI355X_MICROARCH.md)
 
 # Algorithmic HBM bytes per label at lookup-gap 1:
 # ROMix writes N blocks of 128 B once and reads N blocks of 128 B once
-# (random), plus the 16-B label store the metric names.
-BYTES_PER_LABEL = 128 * SCRYPT_N * 2 + 16
+# (random), the working block stages through xbuf twice each way (512 B),
+# plus the 16-B label store the metric names.
+BYTES_PER_LABEL = 128 * SCRYPT_N * 2 + 512 + 16
 
 
 def cpu_baseline():

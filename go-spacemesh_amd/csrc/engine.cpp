@@ -75,9 +75,10 @@ void difficulty_to_be_words(const uint8_t d[32], uint32_t w[8]) {
 constexpr uint32_t THREADS = 256;
 constexpr uint32_t CAND_CAP = 1 << 16;
 
-/* lookup-gap (kernel_args.h): default gap=2 for real scrypt sizes — halves
- * the HBM scratch (2x in-flight labels) and write traffic for ~1 extra
- * BlockMix per phase-2 read at wave level.  POST_GAP_SHIFT overrides. */
+/* lookup-gap (kernel_args.h).  Measured on MI355X (gpurun summary3-5
+ * logs): with the quad-cooperative kernels the scratch fits at gap 1 and
+ * extra recompute only costs, so the default is gap 1 (shift 0).
+ * POST_GAP_SHIFT overrides for scratch-constrained configs. */
 uint32_t pick_gap_shift(uint32_t scrypt_n) {
   const char *env = getenv("POST_GAP_SHIFT");
   if (env) {
@@ -85,7 +86,7 @@ uint32_t pick_gap_shift(uint32_t scrypt_n) {
     while ((scrypt_n >> g) == 0) g--;
     return g;
   }
-  return scrypt_n >= 512 ? 1 : 0;
+  return 0;
 }
 
 struct DeviceTables { /* AES tables resident per device */

@@ -25,13 +25,6 @@
 
 #define POSTE_THREADS 256
 
-/* minimum waves/SIMD forced on the label kernel (register budget): 2 -> 143
- * VGPRs no spills; 4 -> 128 VGPRs with spills confined to the per-label SHA
- * prologue/tail.  Default 2; overridden with -DPOSTE_LABEL_MIN_WAVES=4. */
-#ifndef POSTE_LABEL_MIN_WAVES
-#define POSTE_LABEL_MIN_WAVES 2
-#endif
-
 /* ------------------------- SHA-256 (device) ------------------------- */
 __constant__ uint32_t c_sha_k[64] = {
     0x428a2f98, 0x71374491, 0xb5c0fbcf, 0xe9b5dba5, 0x3956c25b, 0x59f111f1,
