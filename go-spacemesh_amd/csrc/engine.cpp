@@ -256,10 +256,11 @@ int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
   uint64_t range = s->range_end - s->range_start;
   lanes = std::min<uint64_t>(lanes, std::max<uint64_t>(range, 64));
   /* scratch beyond the kernel's resident capacity is wasted: extra
-   * workgroups only queue (grid-stride covers the batch regardless).
-   * resident threads / 4 = resident quads (labels in flight). */
-  lanes = std::min<uint64_t>(lanes, poste_label_kernel_resident_lanes() / 4);
-  lanes = (lanes / 64) * 64; /* 64 quads = one 256-thread workgroup */
+   * workgroups only queue (grid-stride covers the batch regardless) */
+  lanes = std::min<uint64_t>(lanes,
+                             poste_label_resident_slots(s->gap_shift));
+  lanes = (lanes / 128) * 128; /* keep the dual-stream grid block-aligned */
+  if (lanes == 0) lanes = 128;
   if (lanes == 0) {
     delete s;
     set_error("not enough device memory for one scratch lane block");
@@ -912,8 +913,9 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
   uint64_t per_lane = ((uint64_t)cfg->scrypt_n >> gap_shift) * 128;
   uint64_t max_lanes = (uint64_t)((double)free_b * 0.75) / per_lane;
   max_lanes = std::min<uint64_t>(max_lanes,
-                                 poste_label_kernel_resident_lanes() / 4);
-  max_lanes = (max_lanes / 64) * 64;
+                                 poste_label_resident_slots(gap_shift));
+  max_lanes = (max_lanes / 128) * 128;
+  if (max_lanes == 0) max_lanes = 128;
   if (max_lanes == 0) {
     set_error("not enough memory for verification scratch");
     return POST_ERR_OOM;
@@ -926,7 +928,7 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
     h_cid[i] = tasks[i].proof;
   }
   uint64_t lanes = std::min<uint64_t>(
-      max_lanes, ((tasks.size() + 63) / 64) * 64);
+      max_lanes, ((tasks.size() + 127) / 128) * 128);
   uint32_t *d_scratch = nullptr;
   uint64_t *d_idx = nullptr;
   uint32_t *d_cid = nullptr, *d_cm = nullptr;
@@ -1025,7 +1027,7 @@ int post_verify_vrf_nonce(const PostProofMetadata *meta, uint64_t index,
   uint64_t *d_idx = nullptr;
   uint8_t *d_out = nullptr;
   uint32_t *d_xbuf1 = nullptr;
-  HIP_TRY(hipMalloc(&d_scratch, (size_t)64 * per_lane));
+  HIP_TRY(hipMalloc(&d_scratch, (size_t)128 * per_lane));
   HIP_TRY(hipMalloc(&d_idx, 8));
   HIP_TRY(hipMalloc(&d_out, 32));
   HIP_TRY(hipMalloc(&d_xbuf1, 128));
@@ -1038,7 +1040,7 @@ int post_verify_vrf_nonce(const PostProofMetadata *meta, uint64_t index,
   la.xbuf = d_xbuf1;
   la.out_full = 1;
   la.scratch = d_scratch;
-  la.scratch_lanes = 64;
+  la.scratch_lanes = 128;
   la.out = d_out;
   la.indices = d_idx;
   la.count = 1;

@@ -52,7 +52,7 @@ hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
                                      uint32_t blocks, hipStream_t stream);
 hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
                                     uint32_t blocks, hipStream_t stream);
-uint64_t poste_label_kernel_resident_lanes(void);
+uint64_t poste_label_resident_slots(uint32_t gap_shift);
 }
 
 #endif

@@ -124,6 +124,64 @@ __device__ __forceinline__ void salsa8_z(uint32_t &A, uint32_t &B,
   A += a; B += b; C += c; D += d;
 }
 
+/* dual-stream salsa20/8: two independent quads' states interleaved so the
+ * in-order wave always has a second dependency chain to issue from */
+__device__ __forceinline__ void salsa8_z2(uint32_t &A0, uint32_t &B0,
+                                          uint32_t &C0, uint32_t &D0,
+                                          uint32_t &A1, uint32_t &B1,
+                                          uint32_t &C1, uint32_t &D1) {
+  uint32_t a0 = A0, b0 = B0, c0 = C0, d0 = D0;
+  uint32_t a1 = A1, b1 = B1, c1 = C1, d1 = D1;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    b0 ^= __builtin_rotateleft32(a0 + d0, 7);
+    b1 ^= __builtin_rotateleft32(a1 + d1, 7);
+    c0 ^= __builtin_rotateleft32(b0 + a0, 9);
+    c1 ^= __builtin_rotateleft32(b1 + a1, 9);
+    d0 ^= __builtin_rotateleft32(c0 + b0, 13);
+    d1 ^= __builtin_rotateleft32(c1 + b1, 13);
+    a0 ^= __builtin_rotateleft32(d0 + c0, 18);
+    a1 ^= __builtin_rotateleft32(d1 + c1, 18);
+    uint32_t y10 = SWZ(d0, QROT1), y11 = SWZ(d1, QROT1);
+    uint32_t y20 = SWZ(c0, QROT2), y21 = SWZ(c1, QROT2);
+    uint32_t y30 = SWZ(b0, QROT3), y31 = SWZ(b1, QROT3);
+    y10 ^= __builtin_rotateleft32(a0 + y30, 7);
+    y11 ^= __builtin_rotateleft32(a1 + y31, 7);
+    y20 ^= __builtin_rotateleft32(y10 + a0, 9);
+    y21 ^= __builtin_rotateleft32(y11 + a1, 9);
+    y30 ^= __builtin_rotateleft32(y20 + y10, 13);
+    y31 ^= __builtin_rotateleft32(y21 + y11, 13);
+    a0 ^= __builtin_rotateleft32(y30 + y20, 18);
+    a1 ^= __builtin_rotateleft32(y31 + y21, 18);
+    d0 = SWZ(y10, QROT3);
+    d1 = SWZ(y11, QROT3);
+    c0 = SWZ(y20, QROT2);
+    c1 = SWZ(y21, QROT2);
+    b0 = SWZ(y30, QROT1);
+    b1 = SWZ(y31, QROT1);
+  }
+  A0 += a0; B0 += b0; C0 += c0; D0 += d0;
+  A1 += a1; B1 += b1; C1 += c1; D1 += d1;
+}
+
+/* dual-stream BlockMix r=1 */
+__device__ __forceinline__ void blockmix2_z(uint32_t XA0[4], uint32_t XA1[4],
+                                            uint32_t XB0[4],
+                                            uint32_t XB1[4]) {
+  uint32_t ta0 = XA0[0] ^ XA1[0], ta1 = XA0[1] ^ XA1[1],
+           ta2 = XA0[2] ^ XA1[2], ta3 = XA0[3] ^ XA1[3];
+  uint32_t tb0 = XB0[0] ^ XB1[0], tb1 = XB0[1] ^ XB1[1],
+           tb2 = XB0[2] ^ XB1[2], tb3 = XB0[3] ^ XB1[3];
+  salsa8_z2(ta0, ta1, ta2, ta3, tb0, tb1, tb2, tb3);
+  XA0[0] = ta0; XA0[1] = ta1; XA0[2] = ta2; XA0[3] = ta3;
+  XB0[0] = tb0; XB0[1] = tb1; XB0[2] = tb2; XB0[3] = tb3;
+  ta0 ^= XA1[0]; ta1 ^= XA1[1]; ta2 ^= XA1[2]; ta3 ^= XA1[3];
+  tb0 ^= XB1[0]; tb1 ^= XB1[1]; tb2 ^= XB1[2]; tb3 ^= XB1[3];
+  salsa8_z2(ta0, ta1, ta2, ta3, tb0, tb1, tb2, tb3);
+  XA1[0] = ta0; XA1[1] = ta1; XA1[2] = ta2; XA1[3] = ta3;
+  XB1[0] = tb0; XB1[1] = tb1; XB1[2] = tb2; XB1[3] = tb3;
+}
+
 /* BlockMix r=1 on a quad: X = (B0,B1) as two z-vectors per lane (8 regs) */
 __device__ __forceinline__ void blockmix_z(uint32_t X0[4], uint32_t X1[4]) {
   uint32_t t0 = X0[0] ^ X1[0], t1 = X0[1] ^ X1[1], t2 = X0[2] ^ X1[2],
@@ -218,11 +276,13 @@ post_label_prologue_kernel(LabelKernelArgs a) {
   const unsigned long long lane =
       (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
   const unsigned long long group = lane >> 2;
+  const unsigned long long gstride =
+      ((unsigned long long)gridDim.x * blockDim.x) >> 2;
   const uint32_t sub = (uint32_t)(lane & 3);
   uint4 *X = (uint4 *)a.xbuf;
 
   for (unsigned long long task = group; task < a.count;
-       task += a.scratch_lanes) {
+       task += gstride) {
     const unsigned long long index = a.indices ? a.indices[task]
                                                : a.start + task;
     const uint32_t *cw = a.commit_ids
@@ -329,6 +389,87 @@ post_label_romix_kernel(LabelKernelArgs a) {
   }
 }
 
+/* dual-stream ROMix (lookup-gap 1 only): each quad advances TWO labels so
+ * the in-order wave interleaves two independent salsa chains — the
+ * SQ_WAIT_INST_ANY issue-stall fix (profiles/pmc_sq: 54% of wave cycles).
+ * In-flight slots = 2 x grid quads (= a.scratch_lanes). */
+__global__ void __launch_bounds__(POSTE_THREADS, 8)
+post_label_romix2_kernel(LabelKernelArgs a) {
+  const unsigned long long lane =
+      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned long long group = lane >> 2;
+  const uint32_t sub = (uint32_t)(lane & 3);
+  const unsigned long long gquads =
+      ((unsigned long long)gridDim.x * blockDim.x) >> 2;
+  const uint32_t n = a.scrypt_n;
+  const uint32_t mask = n - 1;
+  uint4 *V = (uint4 *)a.scratch;
+  uint4 *X = (uint4 *)a.xbuf;
+  const unsigned long long slotA = group * 2, slotB = group * 2 + 1;
+
+  for (unsigned long long base = group * 2; base < a.count;
+       base += 2 * gquads) {
+    const unsigned long long taskA = base;
+    const unsigned long long taskB = base + 1;
+    const bool validB = taskB < a.count;
+    const unsigned long long tb = validB ? taskB : taskA;
+    uint32_t ZA0[4], ZA1[4], ZB0[4], ZB1[4];
+    {
+      uint4 *pA = X + taskA * 8ull;
+      uint4 *pB = X + tb * 8ull;
+      uint4 a0 = pA[sub], a1 = pA[sub + 4];
+      uint4 b0 = pB[sub], b1 = pB[sub + 4];
+      ZA0[0] = a0.x; ZA0[1] = a0.y; ZA0[2] = a0.z; ZA0[3] = a0.w;
+      ZA1[0] = a1.x; ZA1[1] = a1.y; ZA1[2] = a1.z; ZA1[3] = a1.w;
+      ZB0[0] = b0.x; ZB0[1] = b0.y; ZB0[2] = b0.z; ZB0[3] = b0.w;
+      ZB1[0] = b1.x; ZB1[1] = b1.y; ZB1[2] = b1.z; ZB1[3] = b1.w;
+    }
+    /* phase 1 */
+    {
+      unsigned long long baseA = slotA * 8ull + sub;
+      unsigned long long baseB = slotB * 8ull + sub;
+      const unsigned long long stride = a.scratch_lanes * 8ull;
+      for (uint32_t j = 0; j < n; j++) {
+        uint4 *pA = V + baseA;
+        uint4 *pB = V + baseB;
+        pA[0] = make_uint4(ZA0[0], ZA0[1], ZA0[2], ZA0[3]);
+        pA[4] = make_uint4(ZA1[0], ZA1[1], ZA1[2], ZA1[3]);
+        pB[0] = make_uint4(ZB0[0], ZB0[1], ZB0[2], ZB0[3]);
+        pB[4] = make_uint4(ZB1[0], ZB1[1], ZB1[2], ZB1[3]);
+        blockmix2_z(ZA0, ZA1, ZB0, ZB1);
+        baseA += stride;
+        baseB += stride;
+      }
+    }
+    /* phase 2 */
+    for (uint32_t i = 0; i < n; i++) {
+      uint32_t jA = SWZ(ZA1[0], QBCAST0) & mask;
+      uint32_t jB = SWZ(ZB1[0], QBCAST0) & mask;
+      const uint4 *pA =
+          V + ((unsigned long long)jA * a.scratch_lanes + slotA) * 8ull + sub;
+      const uint4 *pB =
+          V + ((unsigned long long)jB * a.scratch_lanes + slotB) * 8ull + sub;
+      uint4 va0 = pA[0], va1 = pA[4];
+      uint4 vb0 = pB[0], vb1 = pB[4];
+      ZA0[0] ^= va0.x; ZA0[1] ^= va0.y; ZA0[2] ^= va0.z; ZA0[3] ^= va0.w;
+      ZA1[0] ^= va1.x; ZA1[1] ^= va1.y; ZA1[2] ^= va1.z; ZA1[3] ^= va1.w;
+      ZB0[0] ^= vb0.x; ZB0[1] ^= vb0.y; ZB0[2] ^= vb0.z; ZB0[3] ^= vb0.w;
+      ZB1[0] ^= vb1.x; ZB1[1] ^= vb1.y; ZB1[2] ^= vb1.z; ZB1[3] ^= vb1.w;
+      blockmix2_z(ZA0, ZA1, ZB0, ZB1);
+    }
+    {
+      uint4 *pA = X + taskA * 8ull;
+      pA[sub] = make_uint4(ZA0[0], ZA0[1], ZA0[2], ZA0[3]);
+      pA[sub + 4] = make_uint4(ZA1[0], ZA1[1], ZA1[2], ZA1[3]);
+      if (validB) {
+        uint4 *pB = X + taskB * 8ull;
+        pB[sub] = make_uint4(ZB0[0], ZB0[1], ZB0[2], ZB0[3]);
+        pB[sub + 4] = make_uint4(ZB1[0], ZB1[1], ZB1[2], ZB1[3]);
+      }
+    }
+  }
+}
+
 /* tail: PBKDF2(P, X, 1, 32) -> labels, VRF-minimum tracking + exact
  * per-workgroup candidate reduce */
 __global__ void __launch_bounds__(POSTE_THREADS)
@@ -336,6 +477,8 @@ post_label_tail_kernel(LabelKernelArgs a) {
   const unsigned long long lane =
       (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
   const unsigned long long group = lane >> 2;
+  const unsigned long long gstride =
+      ((unsigned long long)gridDim.x * blockDim.x) >> 2;
   const uint32_t sub = (uint32_t)(lane & 3);
   uint4 *X = (uint4 *)a.xbuf;
 
@@ -344,7 +487,7 @@ post_label_tail_kernel(LabelKernelArgs a) {
   int min_found = 0;
 
   for (unsigned long long task = group; task < a.count;
-       task += a.scratch_lanes) {
+       task += gstride) {
     const unsigned long long index = a.indices ? a.indices[task]
                                                : a.start + task;
     const uint32_t *cw = a.commit_ids
@@ -559,32 +702,46 @@ extern "C" {
 
 hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
                                      uint32_t blocks, hipStream_t stream) {
-  hipLaunchKernelGGL(post_label_prologue_kernel, dim3(blocks),
+  /* `blocks` is slots/64.  The dual-stream ROMix covers two slots per quad,
+   * so its grid (and the prologue/tail grids, which stride by their own
+   * size) is half that. */
+  uint32_t gblocks = args->gap_shift == 0 ? (blocks + 1) / 2 : blocks;
+  hipLaunchKernelGGL(post_label_prologue_kernel, dim3(gblocks),
                      dim3(POSTE_THREADS), 0, stream, *args);
-  hipLaunchKernelGGL(post_label_romix_kernel, dim3(blocks),
-                     dim3(POSTE_THREADS), 0, stream, *args);
-  hipLaunchKernelGGL(post_label_tail_kernel, dim3(blocks),
+  if (args->gap_shift == 0) {
+    hipLaunchKernelGGL(post_label_romix2_kernel, dim3(gblocks),
+                       dim3(POSTE_THREADS), 0, stream, *args);
+  } else {
+    hipLaunchKernelGGL(post_label_romix_kernel, dim3(blocks),
+                       dim3(POSTE_THREADS), 0, stream, *args);
+  }
+  hipLaunchKernelGGL(post_label_tail_kernel, dim3(gblocks),
                      dim3(POSTE_THREADS), 0, stream, *args);
   return hipGetLastError();
 }
 
-/* Max simultaneously-resident lanes of the label kernel on the current
- * device (occupancy query x CUs x block size).  Scratch beyond this is
- * wasted memory: extra workgroups just queue. */
-uint64_t poste_label_kernel_resident_lanes(void) {
+/* Max simultaneously-resident label SLOTS (in-flight labels) of the ROMix
+ * kernel for the given gap config.  Scratch beyond this is wasted memory:
+ * extra workgroups just queue. */
+uint64_t poste_label_resident_slots(uint32_t gap_shift) {
+  const void *kern =
+      gap_shift == 0
+          ? reinterpret_cast<const void *>(post_label_romix2_kernel)
+          : reinterpret_cast<const void *>(post_label_romix_kernel);
   int blocks_per_cu = 0;
-  if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
-          &blocks_per_cu,
-          reinterpret_cast<const void *>(post_label_romix_kernel),
-          POSTE_THREADS, 0) != hipSuccess ||
+  if (hipOccupancyMaxActiveBlocksPerMultiprocessor(&blocks_per_cu, kern,
+                                                   POSTE_THREADS, 0) !=
+          hipSuccess ||
       blocks_per_cu <= 0)
     blocks_per_cu = 2;
   hipDeviceProp_t prop;
   int dev = 0;
   (void)hipGetDevice(&dev);
-  if (hipGetDeviceProperties(&prop, dev) != hipSuccess)
-    return (uint64_t)blocks_per_cu * 256 * POSTE_THREADS;
-  return (uint64_t)blocks_per_cu * prop.multiProcessorCount * POSTE_THREADS;
+  int cus = 256;
+  if (hipGetDeviceProperties(&prop, dev) == hipSuccess)
+    cus = prop.multiProcessorCount;
+  uint64_t quads = (uint64_t)blocks_per_cu * cus * (POSTE_THREADS / 4);
+  return gap_shift == 0 ? quads * 2 : quads;
 }
 
 hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
