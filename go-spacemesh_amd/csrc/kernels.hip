@@ -20,6 +20,8 @@
  */
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+
 #include "kernel_args.h"
 #include "post_common.h"
 
@@ -702,13 +704,18 @@ extern "C" {
 
 hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
                                      uint32_t blocks, hipStream_t stream) {
-  /* `blocks` is slots/64.  The dual-stream ROMix covers two slots per quad,
-   * so its grid (and the prologue/tail grids, which stride by their own
-   * size) is half that. */
-  uint32_t gblocks = args->gap_shift == 0 ? (blocks + 1) / 2 : blocks;
+  /* Dual-stream ROMix (two slots per quad, half the grid) measured
+   * neutral-to-slightly-slower than single-stream at 8 waves/SIMD
+   * (gpurun summary6); kept behind POST_ROMIX2=1 for re-evaluation. */
+  static const bool use2 = [] {
+    const char *e = getenv("POST_ROMIX2");
+    return e && e[0] == '1';
+  }();
+  const bool dual = use2 && args->gap_shift == 0;
+  uint32_t gblocks = dual ? (blocks + 1) / 2 : blocks;
   hipLaunchKernelGGL(post_label_prologue_kernel, dim3(gblocks),
                      dim3(POSTE_THREADS), 0, stream, *args);
-  if (args->gap_shift == 0) {
+  if (dual) {
     hipLaunchKernelGGL(post_label_romix2_kernel, dim3(gblocks),
                        dim3(POSTE_THREADS), 0, stream, *args);
   } else {
@@ -741,7 +748,9 @@ uint64_t poste_label_resident_slots(uint32_t gap_shift) {
   if (hipGetDeviceProperties(&prop, dev) == hipSuccess)
     cus = prop.multiProcessorCount;
   uint64_t quads = (uint64_t)blocks_per_cu * cus * (POSTE_THREADS / 4);
-  return gap_shift == 0 ? quads * 2 : quads;
+  const char *e = getenv("POST_ROMIX2");
+  const bool dual = e && e[0] == '1' && gap_shift == 0;
+  return dual ? quads * 2 : quads;
 }
 
 hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
