@@ -257,3 +257,31 @@ def test_providers_and_benchmark():
     lps = eng.benchmark(0, 8192)
     assert lps > 0
     print("benchmark labels/s:", lps)
+
+
+def test_prove_from_datadir_roundtrip(tmp_path):
+    """File-backed proving: init to postdata_*.bin, prove from the
+    directory (the post-service role over a real data dir), verify."""
+    d = str(tmp_path)
+    NU, LPU, N = 1, 1 << 11, 32
+    cfg, mgr = make_mgr(NU, LPU, N, data_dir=d,
+                        max_file_size=(1 << 10) * 16)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    mgr.reset()
+    pcfg = gsm_amd.PostConfig(min_num_units=1, labels_per_unit=LPU, k1=12,
+                              k2=8, k3=8, pow_difficulty=POW_DIFF)
+    proof = gsm_amd.api.prove_dir(d, CHALLENGE, pcfg,
+                                  gsm_amd.ProveOpts(nonces=16))
+    ver = gsm_amd.PostVerifier(pcfg, scrypt_n=N)
+    meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, NU, LPU)
+    ver.verify(proof, meta)
+    # and the oracle agrees with the file-backed proof
+    labels = b"".join(
+        open(os.path.join(d, f), "rb").read()
+        for f in sorted(os.listdir(d))
+        if f.startswith("postdata_") and f.endswith(".bin"))
+    o = Oracle()
+    op = o.prove(labels, NU * LPU, CHALLENGE, 12, 8, 16, POW_DIFF)
+    assert proof.nonce == op.nonce
+    assert proof.indices == bytes(op.indices[:op.indices_len])
