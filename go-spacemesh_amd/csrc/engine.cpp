@@ -231,14 +231,19 @@ int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
   int rc = require_gpu(cfg->provider_id);
   if (rc != POST_OK) return rc;
 
+  /* everything below must free the session on failure */
+  struct Guard {
+    PostInitSession *p;
+    ~Guard() { delete p; }
+  };
   auto *s = new PostInitSession();
+  Guard guard{s};
   s->cfg = *cfg;
   if (cfg->data_dir) s->data_dir = cfg->data_dir;
   uint64_t total = (uint64_t)cfg->num_units * cfg->labels_per_unit;
   s->range_start = cfg->index_start;
   s->range_end = cfg->index_end ? cfg->index_end : total;
   if (s->range_end > total || s->range_start >= s->range_end) {
-    delete s;
     set_error("bad index range");
     return POST_ERR_INVALID_ARGS;
   }
@@ -262,7 +267,6 @@ int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
   lanes = (lanes / 128) * 128; /* keep the dual-stream grid block-aligned */
   if (lanes == 0) lanes = 128;
   if (lanes == 0) {
-    delete s;
     set_error("not enough device memory for one scratch lane block");
     return POST_ERR_OOM;
   }
@@ -287,6 +291,7 @@ int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
         1, s->cfg.max_file_size / POST_LABEL_SIZE);
     s->written = existing_labels(s->data_dir, per_file, s->range_start, range);
   }
+  guard.p = nullptr; /* ownership passes to the caller */
   *out = s;
   return POST_OK;
 }

@@ -285,3 +285,67 @@ def test_prove_from_datadir_roundtrip(tmp_path):
     op = o.prove(labels, NU * LPU, CHALLENGE, 12, 8, 16, POW_DIFF)
     assert proof.nonce == op.nonce
     assert proof.indices == bytes(op.indices[:op.indices_len])
+
+
+def test_high_index_labels_golden():
+    """Label indices above 2^32 exercise the high word of LE64(index) in
+    the device PBKDF2 password — pinned by the OpenSSL golden vector at
+    index 2^32+17 (N=8192)."""
+    hi = next(v for v in GOLDEN["labels_openssl"]["labels"]
+              if v["N"] == 8192 and v["index"] > 2**32)
+    start = 2**32
+    cfg, mgr = make_mgr(2, 2**32, 8192, index_start=start,
+                        index_end=start + 64)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    got = mgr.copy_labels(0, 64)
+    off = hi["index"] - start
+    assert got[off * 16:(off + 1) * 16].hex() == hi["full"][:32]
+    mgr.reset()
+
+
+def test_ragged_range_parity():
+    """Non-power-of-two, non-wave-aligned batch sizes vs the oracle."""
+    total = 1000  # not a multiple of 64 quads or 128 slots
+    cfg, mgr = make_mgr(1, total, 128)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    got = mgr.copy_labels(0, total)
+    o = Oracle()
+    commit = o.commitment(NODE, ATX)
+    want, best = o.init_range(commit, 0, total, 128)
+    assert got == want
+    nonce = mgr.vrf_nonce()
+    assert nonce is not None and nonce[0] == best.index
+    mgr.reset()
+
+
+def test_shard_union_equals_whole(tmp_path):
+    """Three shard sessions over [0,T) produce byte-identical labels to a
+    single whole-range session, and the merged nonce equals the whole-range
+    nonce (the 8-GPU sharding axis)."""
+    import importlib
+    sharding = importlib.import_module("go-spacemesh_amd.sharding")
+    T, N = 3000, 128
+    whole_cfg, whole = make_mgr(1, T, N)
+    whole.prepare_initializer()
+    whole.start_session()
+    ref_labels = whole.copy_labels(0, T)
+    ref_nonce = whole.vrf_nonce()
+    whole.reset()
+
+    parts = []
+    cands = []
+    for r in range(3):
+        s, e = sharding.shard_range(T, 3, r)
+        cfg, mgr = make_mgr(1, T, N, index_start=s, index_end=e)
+        mgr.prepare_initializer()
+        mgr.start_session()
+        parts.append(mgr.copy_labels(0, e - s))
+        n = mgr.vrf_nonce()
+        cands.append((n[0], n[1]) if n else None)
+        mgr.reset()
+    assert b"".join(parts) == ref_labels
+    merged = sharding.merge_nonces(cands)
+    assert merged is not None and ref_nonce is not None
+    assert merged[0] == ref_nonce[0]
