@@ -89,6 +89,69 @@ uint32_t pick_gap_shift(uint32_t scrypt_n) {
   return 0;
 }
 
+/* Cached per-device verification workspace: the worker-pool call pattern
+ * (one proof per post_verify call, post_verifier.go:150-160) would
+ * otherwise pay scratch/staging allocation on every call.  Calls remain
+ * safe for concurrent workers; they serialize on the device section. */
+struct VerifyWorkspace {
+  std::mutex mu;
+  uint32_t *d_scratch = nullptr;
+  size_t scratch_cap = 0; /* bytes */
+  uint64_t *d_idx = nullptr;
+  uint32_t *d_cid = nullptr;
+  uint8_t *d_out = nullptr;
+  uint32_t *d_xbuf = nullptr;
+  size_t tasks_cap = 0;
+  uint32_t *d_cm = nullptr;
+  size_t cm_cap = 0; /* words */
+};
+std::mutex g_vws_mu;
+std::map<int, VerifyWorkspace *> g_vws;
+
+VerifyWorkspace *get_verify_ws(int dev) {
+  std::lock_guard<std::mutex> lk(g_vws_mu);
+  auto it = g_vws.find(dev);
+  if (it != g_vws.end()) return it->second;
+  auto *w = new VerifyWorkspace();
+  g_vws[dev] = w;
+  return w;
+}
+
+int ws_reserve(VerifyWorkspace *w, size_t scratch_bytes, size_t tasks,
+               size_t cm_words) {
+  if (scratch_bytes > w->scratch_cap) {
+    if (w->d_scratch) (void)hipFree(w->d_scratch);
+    w->d_scratch = nullptr;
+    w->scratch_cap = 0;
+    HIP_TRY(hipMalloc(&w->d_scratch, scratch_bytes));
+    w->scratch_cap = scratch_bytes;
+  }
+  if (tasks > w->tasks_cap) {
+    size_t cap = std::max<size_t>(tasks, 4096);
+    if (w->d_idx) (void)hipFree(w->d_idx);
+    if (w->d_cid) (void)hipFree(w->d_cid);
+    if (w->d_out) (void)hipFree(w->d_out);
+    if (w->d_xbuf) (void)hipFree(w->d_xbuf);
+    w->d_idx = nullptr; w->d_cid = nullptr; w->d_out = nullptr;
+    w->d_xbuf = nullptr;
+    w->tasks_cap = 0;
+    HIP_TRY(hipMalloc(&w->d_idx, cap * 8));
+    HIP_TRY(hipMalloc(&w->d_cid, cap * 4));
+    HIP_TRY(hipMalloc(&w->d_out, cap * 32));
+    HIP_TRY(hipMalloc(&w->d_xbuf, cap * 128));
+    w->tasks_cap = cap;
+  }
+  if (cm_words > w->cm_cap) {
+    size_t cap = std::max<size_t>(cm_words, 1024);
+    if (w->d_cm) (void)hipFree(w->d_cm);
+    w->d_cm = nullptr;
+    w->cm_cap = 0;
+    HIP_TRY(hipMalloc(&w->d_cm, cap * 4));
+    w->cm_cap = cap;
+  }
+  return POST_OK;
+}
+
 struct DeviceTables { /* AES tables resident per device */
   uint32_t *d_te = nullptr;
   uint8_t *d_sbox = nullptr;
@@ -934,48 +997,40 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
   }
   uint64_t lanes = std::min<uint64_t>(
       max_lanes, ((tasks.size() + 127) / 128) * 128);
-  uint32_t *d_scratch = nullptr;
-  uint64_t *d_idx = nullptr;
-  uint32_t *d_cid = nullptr, *d_cm = nullptr;
-  uint8_t *d_out = nullptr;
-  uint32_t *d_xbuf = nullptr;
-  HIP_TRY(hipMalloc(&d_scratch, (size_t)lanes * per_lane));
-  HIP_TRY(hipMalloc(&d_xbuf, (size_t)tasks.size() * 128));
-  HIP_TRY(hipMalloc(&d_idx, h_idx.size() * 8));
-  HIP_TRY(hipMalloc(&d_cid, h_cid.size() * 4));
-  HIP_TRY(hipMalloc(&d_cm, commit_words.size() * 4));
-  HIP_TRY(hipMalloc(&d_out, tasks.size() * 32));
-  HIP_TRY(hipMemcpy(d_idx, h_idx.data(), h_idx.size() * 8,
-                    hipMemcpyHostToDevice));
-  HIP_TRY(hipMemcpy(d_cid, h_cid.data(), h_cid.size() * 4,
-                    hipMemcpyHostToDevice));
-  HIP_TRY(hipMemcpy(d_cm, commit_words.data(), commit_words.size() * 4,
-                    hipMemcpyHostToDevice));
 
-  LabelKernelArgs la;
-  std::memset(&la, 0, sizeof(la));
-  la.scrypt_n = cfg->scrypt_n;
-  la.gap_shift = gap_shift;
-  la.xbuf = d_xbuf;
-  la.out_full = 1;
-  la.scratch = d_scratch;
-  la.scratch_lanes = lanes;
-  la.out = d_out;
-  la.indices = h_idx.size() ? d_idx : nullptr;
-  la.commit_ids = d_cid;
-  la.commitments = d_cm;
-  la.count = tasks.size();
-  HIP_TRY(poste_launch_label_kernel(&la, (uint32_t)(lanes / 64), nullptr));
-  HIP_TRY(hipDeviceSynchronize());
   std::vector<uint8_t> full((size_t)tasks.size() * 32);
-  HIP_TRY(hipMemcpy(full.data(), d_out, full.size(),
-                    hipMemcpyDeviceToHost));
-  (void)hipFree(d_scratch);
-  (void)hipFree(d_xbuf);
-  (void)hipFree(d_idx);
-  (void)hipFree(d_cid);
-  (void)hipFree(d_cm);
-  (void)hipFree(d_out);
+  {
+    VerifyWorkspace *ws = get_verify_ws((int)cfg->provider_id);
+    std::lock_guard<std::mutex> lk(ws->mu);
+    rc = ws_reserve(ws, (size_t)lanes * per_lane, tasks.size(),
+                    commit_words.size());
+    if (rc != POST_OK) return rc;
+    HIP_TRY(hipMemcpy(ws->d_idx, h_idx.data(), h_idx.size() * 8,
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(ws->d_cid, h_cid.data(), h_cid.size() * 4,
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(ws->d_cm, commit_words.data(),
+                      commit_words.size() * 4, hipMemcpyHostToDevice));
+
+    LabelKernelArgs la;
+    std::memset(&la, 0, sizeof(la));
+    la.scrypt_n = cfg->scrypt_n;
+    la.gap_shift = gap_shift;
+    la.xbuf = ws->d_xbuf;
+    la.out_full = 1;
+    la.scratch = ws->d_scratch;
+    la.scratch_lanes = lanes;
+    la.out = ws->d_out;
+    la.indices = h_idx.size() ? ws->d_idx : nullptr;
+    la.commit_ids = ws->d_cid;
+    la.commitments = ws->d_cm;
+    la.count = tasks.size();
+    HIP_TRY(poste_launch_label_kernel(&la, (uint32_t)(lanes / 64),
+                                      nullptr));
+    HIP_TRY(hipDeviceSynchronize());
+    HIP_TRY(hipMemcpy(full.data(), ws->d_out, full.size(),
+                      hipMemcpyDeviceToHost));
+  }
 
   /* final AES threshold predicate on host (cheap; K3*n blocks) */
   std::vector<std::array<uint32_t, 44>> rk_cache(n);
