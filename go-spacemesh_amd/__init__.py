@@ -11,6 +11,8 @@ There is NO CPU fallback here: on a machine with a GPU the HIP engine must
 load and run, and missing kernels raise immediately.  The CPU oracle under
 oracle/ is test infrastructure only and is never imported by this package.
 """
+from . import events  # noqa: F401
+from .verifier_pool import OffloadingVerifier  # noqa: F401
 from .api import (  # noqa: F401
     Engine,
     EngineError,
@@ -26,7 +28,7 @@ from .api import (  # noqa: F401
 )
 
 __all__ = [
-    "Engine", "EngineError", "PostConfig", "PostProof", "PostProofMetadata",
-    "PostSetupManager", "PostSetupOpts", "PostVerifier", "ProveOpts",
-    "VerifyOpts", "load_engine",
+    "Engine", "EngineError", "OffloadingVerifier", "PostConfig", "PostProof",
+    "PostProofMetadata", "PostSetupManager", "PostSetupOpts", "PostVerifier",
+    "ProveOpts", "VerifyOpts", "events", "load_engine",
 ]

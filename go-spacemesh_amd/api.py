@@ -22,6 +22,8 @@ from ctypes import (POINTER, byref, c_char_p, c_int, c_int32, c_size_t,
                     c_uint8, c_uint16, c_uint32, c_uint64, c_void_p)
 from typing import Optional
 
+from . import events as _ev
+
 _DIR = os.path.dirname(os.path.abspath(__file__))
 _LIB_PATH = os.environ.get("POST_ENGINE_LIB",
                            os.path.join(_DIR, "libpost_hip.so"))
@@ -334,14 +336,18 @@ class PostSetupManager:
         if self.state != self.PREPARED:
             raise EngineError(Status.ERR, "post session not prepared")
         self.state = self.IN_PROGRESS
+        _ev.bus().emit(_ev.InitStart(self.node_id, self.commitment_atx_id))
         rc = self.engine.lib.post_init_run(self._session)
         if rc == Status.CANCELLED:
             self.state = self.STOPPED
             raise EngineError(rc, "stopped")
         if rc != 0:
             self.state = self.ERROR
-            raise EngineError(rc, self.engine.lib.post_last_error().decode())
+            err = self.engine.lib.post_last_error().decode()
+            _ev.bus().emit(_ev.InitFailure(self.node_id, err))
+            raise EngineError(rc, err)
         self.state = self.COMPLETE
+        _ev.bus().emit(_ev.InitComplete(self.node_id))
 
     def step(self, max_labels: int) -> tuple:
         """Process up to max_labels labels; returns (labels_done,
@@ -394,6 +400,17 @@ class PostSetupManager:
 def prove_buffer(labels: bytes, num_labels: int, node_id: bytes,
                  atx_id: bytes, challenge: bytes, cfg: PostConfig,
                  opts: ProveOpts) -> PostProof:
+    _ev.bus().emit(_ev.PostStart(node_id, challenge))
+    try:
+        return _prove_buffer(labels, num_labels, node_id, atx_id, challenge,
+                             cfg, opts)
+    finally:
+        _ev.bus().emit(_ev.PostComplete(node_id))
+
+
+def _prove_buffer(labels: bytes, num_labels: int, node_id: bytes,
+                  atx_id: bytes, challenge: bytes, cfg: PostConfig,
+                  opts: ProveOpts) -> PostProof:
     eng = Engine()
     pc = _CProveConfig()
     ctypes.memmove(pc.challenge, challenge, 32)
