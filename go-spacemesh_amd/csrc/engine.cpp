@@ -17,6 +17,7 @@
 #include <algorithm>
 #include <array>
 #include <atomic>
+#include <chrono>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
@@ -999,12 +1000,24 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
       max_lanes, ((tasks.size() + 127) / 128) * 128);
 
   std::vector<uint8_t> full((size_t)tasks.size() * 32);
+  const bool dbg = getenv("POST_VERIFY_DEBUG") != nullptr;
+  auto tick = [] {
+    return std::chrono::duration<double>(
+               std::chrono::steady_clock::now().time_since_epoch())
+        .count();
+  };
+  double t0 = tick();
   {
     VerifyWorkspace *ws = get_verify_ws((int)cfg->provider_id);
     std::lock_guard<std::mutex> lk(ws->mu);
     rc = ws_reserve(ws, (size_t)lanes * per_lane, tasks.size(),
                     commit_words.size());
     if (rc != POST_OK) return rc;
+    if (dbg) {
+      std::fprintf(stderr, "[verify] reserve %.3fs (tasks=%zu lanes=%llu)\n",
+                   tick() - t0, tasks.size(), (unsigned long long)lanes);
+      t0 = tick();
+    }
     HIP_TRY(hipMemcpy(ws->d_idx, h_idx.data(), h_idx.size() * 8,
                       hipMemcpyHostToDevice));
     HIP_TRY(hipMemcpy(ws->d_cid, h_cid.data(), h_cid.size() * 4,
@@ -1025,9 +1038,17 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
     la.commit_ids = ws->d_cid;
     la.commitments = ws->d_cm;
     la.count = tasks.size();
+    if (dbg) {
+      std::fprintf(stderr, "[verify] h2d %.3fs\n", tick() - t0);
+      t0 = tick();
+    }
     HIP_TRY(poste_launch_label_kernel(&la, (uint32_t)(lanes / 64),
                                       nullptr));
     HIP_TRY(hipDeviceSynchronize());
+    if (dbg) {
+      std::fprintf(stderr, "[verify] kernel %.3fs\n", tick() - t0);
+      t0 = tick();
+    }
     HIP_TRY(hipMemcpy(full.data(), ws->d_out, full.size(),
                       hipMemcpyDeviceToHost));
   }
