@@ -14,6 +14,9 @@
 
 #include <hip/hip_runtime.h>
 
+#include <sys/stat.h>
+#include <sys/types.h>
+
 #include <algorithm>
 #include <array>
 #include <atomic>
@@ -303,7 +306,22 @@ int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
   auto *s = new PostInitSession();
   Guard guard{s};
   s->cfg = *cfg;
-  if (cfg->data_dir) s->data_dir = cfg->data_dir;
+  if (cfg->data_dir) {
+    s->data_dir = cfg->data_dir;
+    /* create the data directory like the reference initializer does */
+    std::string path;
+    for (size_t i = 0; i <= s->data_dir.size(); i++) {
+      if (i == s->data_dir.size() || s->data_dir[i] == '/') {
+        path = s->data_dir.substr(0, i);
+        if (!path.empty()) (void)::mkdir(path.c_str(), 0755);
+      }
+    }
+    struct stat st;
+    if (stat(s->data_dir.c_str(), &st) != 0 || !S_ISDIR(st.st_mode)) {
+      set_error("cannot create data dir " + s->data_dir);
+      return POST_ERR_IO;
+    }
+  }
   uint64_t total = (uint64_t)cfg->num_units * cfg->labels_per_unit;
   s->range_start = cfg->index_start;
   s->range_end = cfg->index_end ? cfg->index_end : total;
