@@ -24,6 +24,7 @@
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <functional>
 #include <map>
 #include <mutex>
 #include <string>
@@ -752,9 +753,12 @@ int post_benchmark(uint32_t provider_id, uint32_t scrypt_n,
 
 /* ------------------------- proving ------------------------- */
 
-static int prove_core(const uint8_t *host_labels, /* chunked source */
-                      uint64_t num_labels, const PostProveConfig *cfg,
-                      PostProof *out) {
+/* label source: fill dst with labels [base, base+cnt) (16 B each) */
+using LabelReader =
+    std::function<int(uint64_t base, uint64_t cnt, uint8_t *dst)>;
+
+static int prove_core(const LabelReader &read_labels, uint64_t num_labels,
+                      const PostProveConfig *cfg, PostProof *out) {
   if (cfg->nonces == 0 || cfg->nonces % POSTE_NONCE_GROUP != 0) {
     set_error("nonces must be a positive multiple of 16");
     return POST_ERR_INVALID_ARGS;
@@ -789,14 +793,50 @@ static int prove_core(const uint8_t *host_labels, /* chunked source */
   HIP_TRY(hipMalloc(&d_rk, rk.size() * 4));
   HIP_TRY(hipMemcpy(d_rk, rk.data(), rk.size() * 4, hipMemcpyHostToDevice));
 
+  /* double-buffered pipeline: disk/host read of chunk i+1 overlaps the
+   * device scan of chunk i (the 256-GiB config-4 pass never needs the
+   * whole label set in host memory).  Expected total hits = nonces*k1
+   * (a few thousand) regardless of label count, so hits accumulate in one
+   * device buffer drained once at the end. */
   const uint64_t CHUNK = 1ull << 24; /* 16M labels = 256 MiB per chunk */
-  uint8_t *d_labels = nullptr;
-  HIP_TRY(hipMalloc(&d_labels, std::min(CHUNK, num_labels) * 16));
+  const uint64_t chunk_lab = std::min(CHUNK, num_labels);
+  uint8_t *d_labels[2] = {nullptr, nullptr};
+  uint8_t *h_labels[2] = {nullptr, nullptr};
+  hipStream_t stream = nullptr;
+  hipEvent_t done[2] = {nullptr, nullptr};
   const uint32_t HIT_CAP = 1u << 22;
   PostScanHit *d_hits = nullptr;
   unsigned int *d_hit_count = nullptr;
+  auto cleanup = [&] {
+    for (int p = 0; p < 2; p++) {
+      if (d_labels[p]) (void)hipFree(d_labels[p]);
+      if (h_labels[p]) (void)hipHostFree(h_labels[p]);
+      if (done[p]) (void)hipEventDestroy(done[p]);
+    }
+    if (stream) (void)hipStreamDestroy(stream);
+    if (d_hits) (void)hipFree(d_hits);
+    if (d_hit_count) (void)hipFree(d_hit_count);
+    (void)hipFree(d_rk);
+  };
+#undef HIP_TRY
+#define HIP_TRY(expr)                                                          \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      cleanup();                                                               \
+      return hip_fail(#expr, _e);                                              \
+    }                                                                          \
+  } while (0)
+  for (int p = 0; p < 2; p++) {
+    HIP_TRY(hipMalloc(&d_labels[p], chunk_lab * 16));
+    HIP_TRY(hipHostMalloc(&h_labels[p], chunk_lab * 16));
+    HIP_TRY(hipEventCreate(&done[p]));
+  }
+  HIP_TRY(hipStreamCreate(&stream));
   HIP_TRY(hipMalloc(&d_hits, sizeof(PostScanHit) * HIT_CAP));
   HIP_TRY(hipMalloc(&d_hit_count, 4));
+  unsigned int zero = 0;
+  HIP_TRY(hipMemcpy(d_hit_count, &zero, 4, hipMemcpyHostToDevice));
 
   ScanKernelArgs sa;
   std::memset(&sa, 0, sizeof(sa));
@@ -809,33 +849,43 @@ static int prove_core(const uint8_t *host_labels, /* chunked source */
   sa.hit_count = d_hit_count;
   sa.hit_cap = HIT_CAP;
 
-  std::vector<PostScanHit> all_hits;
-  for (uint64_t base = 0; base < num_labels; base += CHUNK) {
+  int parity = 0;
+  bool inflight[2] = {false, false};
+  for (uint64_t base = 0; base < num_labels; base += CHUNK, parity ^= 1) {
     uint64_t cnt = std::min(CHUNK, num_labels - base);
-    HIP_TRY(hipMemcpy(d_labels, host_labels + base * 16, cnt * 16,
-                      hipMemcpyHostToDevice));
-    unsigned int zero = 0;
-    HIP_TRY(hipMemcpy(d_hit_count, &zero, 4, hipMemcpyHostToDevice));
-    sa.labels = (const uint4 *)d_labels;
+    /* wait until this parity's buffers are free, then stage the chunk */
+    if (inflight[parity]) HIP_TRY(hipEventSynchronize(done[parity]));
+    int rrc = read_labels(base, cnt, h_labels[parity]);
+    if (rrc != POST_OK) {
+      cleanup();
+      return rrc;
+    }
+    HIP_TRY(hipMemcpyAsync(d_labels[parity], h_labels[parity], cnt * 16,
+                           hipMemcpyHostToDevice, stream));
+    sa.labels = (const uint4 *)d_labels[parity];
     sa.count = cnt;
     sa.index_base = base;
     uint32_t blocks = (uint32_t)std::min<uint64_t>(
         (cnt + THREADS - 1) / THREADS, 8192);
-    HIP_TRY(poste_launch_scan_kernel(&sa, blocks, nullptr));
-    HIP_TRY(hipDeviceSynchronize());
-    unsigned int n_hits = 0;
-    HIP_TRY(hipMemcpy(&n_hits, d_hit_count, 4, hipMemcpyDeviceToHost));
-    if (n_hits > HIT_CAP) n_hits = HIT_CAP;
-    size_t off = all_hits.size();
-    all_hits.resize(off + n_hits);
-    if (n_hits)
-      HIP_TRY(hipMemcpy(all_hits.data() + off, d_hits,
-                        sizeof(PostScanHit) * n_hits, hipMemcpyDeviceToHost));
+    HIP_TRY(poste_launch_scan_kernel(&sa, blocks, stream));
+    HIP_TRY(hipEventRecord(done[parity], stream));
+    inflight[parity] = true;
   }
-  (void)hipFree(d_labels);
-  (void)hipFree(d_hits);
-  (void)hipFree(d_hit_count);
-  (void)hipFree(d_rk);
+  HIP_TRY(hipStreamSynchronize(stream));
+  unsigned int n_hits = 0;
+  HIP_TRY(hipMemcpy(&n_hits, d_hit_count, 4, hipMemcpyDeviceToHost));
+  if (n_hits > HIT_CAP) n_hits = HIT_CAP;
+  std::vector<PostScanHit> all_hits(n_hits);
+  if (n_hits)
+    HIP_TRY(hipMemcpy(all_hits.data(), d_hits, sizeof(PostScanHit) * n_hits,
+                      hipMemcpyDeviceToHost));
+#undef HIP_TRY
+#define HIP_TRY(expr)                                                          \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) return hip_fail(#expr, _e);                          \
+  } while (0)
+  cleanup();
 
   /* winner: nonce whose k2-th smallest passing index is smallest
    * (streaming-order first across the ascending scan); tie -> lowest nonce */
@@ -878,14 +928,25 @@ int post_prove_buffer(const uint8_t *labels, uint64_t num_labels,
   (void)node_id;
   (void)commitment_atx_id; /* scan operates on labels only */
   if (!labels || !cfg || !out) return POST_ERR_INVALID_ARGS;
-  return prove_core(labels, num_labels, cfg, out);
+  LabelReader reader = [labels](uint64_t base, uint64_t cnt, uint8_t *dst) {
+    std::memcpy(dst, labels + base * 16, cnt * 16);
+    return POST_OK;
+  };
+  return prove_core(reader, num_labels, cfg, out);
 }
 
 int post_prove(const char *data_dir, const PostProveConfig *cfg,
                PostProof *out) {
   if (!data_dir || !cfg || !out) return POST_ERR_INVALID_ARGS;
-  /* concatenate postdata_*.bin in index order */
-  std::vector<uint8_t> labels;
+  /* map postdata_*.bin in index order; stream chunks (never the whole set
+   * in host memory — config 4 is 256 GiB) */
+  struct FileSpan {
+    std::string path;
+    uint64_t first_label;
+    uint64_t labels;
+  };
+  std::vector<FileSpan> files;
+  uint64_t total = 0;
   for (uint64_t i = 0;; i++) {
     char path[4096];
     std::snprintf(path, sizeof path, "%s/postdata_%llu.bin", data_dir,
@@ -894,21 +955,46 @@ int post_prove(const char *data_dir, const PostProveConfig *cfg,
     if (!f) break;
     std::fseek(f, 0, SEEK_END);
     long sz = std::ftell(f);
-    std::fseek(f, 0, SEEK_SET);
-    size_t off = labels.size();
-    labels.resize(off + (size_t)sz);
-    if (std::fread(labels.data() + off, 1, (size_t)sz, f) != (size_t)sz) {
-      std::fclose(f);
-      set_error("short read");
-      return POST_ERR_IO;
-    }
     std::fclose(f);
+    uint64_t nlab = (uint64_t)(sz < 0 ? 0 : sz) / 16;
+    files.push_back({path, total, nlab});
+    total += nlab;
   }
-  if (labels.empty()) {
+  if (total == 0) {
     set_error("no postdata_*.bin in data_dir");
     return POST_ERR_IO;
   }
-  return prove_core(labels.data(), labels.size() / 16, cfg, out);
+  LabelReader reader = [&files](uint64_t base, uint64_t cnt, uint8_t *dst)
+      -> int {
+    uint64_t done = 0;
+    for (const auto &fs : files) {
+      if (done == cnt) break;
+      uint64_t want = base + done;
+      if (want < fs.first_label || want >= fs.first_label + fs.labels)
+        continue;
+      uint64_t in_file = want - fs.first_label;
+      uint64_t take = std::min(fs.labels - in_file, cnt - done);
+      FILE *f = std::fopen(fs.path.c_str(), "rb");
+      if (!f) {
+        set_error("cannot reopen " + fs.path);
+        return POST_ERR_IO;
+      }
+      if (std::fseek(f, (long)(in_file * 16), SEEK_SET) != 0 ||
+          std::fread(dst + done * 16, 16, take, f) != take) {
+        std::fclose(f);
+        set_error("short read from " + fs.path);
+        return POST_ERR_IO;
+      }
+      std::fclose(f);
+      done += take;
+    }
+    if (done != cnt) {
+      set_error("label range not covered by postdata files");
+      return POST_ERR_IO;
+    }
+    return POST_OK;
+  };
+  return prove_core(reader, total, cfg, out);
 }
 
 /* ------------------------- verification ------------------------- */
