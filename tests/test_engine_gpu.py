@@ -349,3 +349,41 @@ def test_shard_union_equals_whole(tmp_path):
     merged = sharding.merge_nonces(cands)
     assert merged is not None and ref_nonce is not None
     assert merged[0] == ref_nonce[0]
+
+
+def test_random_far_ranges_match_oracle():
+    """Size-independent property at BASELINE's full index space: arbitrary
+    far shard ranges (up to 2^36) produce exactly the oracle's labels.
+    The full 256-GiB space can't be held, but label i depends only on
+    (commitment, i), so spot ranges pin the whole space."""
+    import random
+    rng = random.Random(31)
+    o = Oracle()
+    commit = o.commitment(NODE, ATX)
+    total_units, lpu = 16, 1 << 32  # 2^36 labels, SURVEY cfg2/3 scale
+    for _ in range(3):
+        start = rng.randrange(0, total_units * lpu - 256)
+        cfg, mgr = make_mgr(total_units, lpu, 8192, index_start=start,
+                            index_end=start + 256)
+        mgr.prepare_initializer()
+        mgr.start_session()
+        got = mgr.copy_labels(0, 256)
+        mgr.reset()
+        for off in (0, 97, 255):
+            want = o.label(commit, start + off, 8192)[:16]
+            assert got[off * 16:(off + 1) * 16] == want, (start, off)
+
+
+def test_verify_malformed_shapes():
+    """Shape checks precede GPU work: wrong indices length and k2 mismatch
+    are INVALID_ARGS, not crashes (validation of the wire cap and
+    bits-per-index, wire_v1.go:43)."""
+    cfg = gsm_amd.PostConfig(min_num_units=1, labels_per_unit=1 << 10,
+                             k1=12, k2=8, k3=8, pow_difficulty=POW_DIFF)
+    ver = gsm_amd.PostVerifier(cfg, scrypt_n=32)
+    meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, 1, 1 << 10)
+    for bad_indices in [b"", b"\x00" * 3, b"\x00" * 100]:
+        with pytest.raises(gsm_amd.EngineError) as ei:
+            ver.verify(gsm_amd.PostProof(0, bad_indices, 0), meta)
+        assert ei.value.code in (gsm_amd.api.Status.INVALID_ARGS,
+                                 gsm_amd.api.Status.POW)

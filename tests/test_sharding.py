@@ -41,6 +41,11 @@ def _gloo_worker(rank, world, tmpdir):
     got = sharding.allreduce_nonce(local)
     # rank 0's label (0x01...) is the global minimum
     assert got == (1, bytes([1]) + bytes(31)), got
+    # the bench's whole-job MAX-over-ranks reduction (bench.py timed region)
+    import torch
+    t = torch.tensor([float(rank + 1)], dtype=torch.float64)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    assert t.item() == world
     dist.barrier()
     dist.destroy_process_group()
 
