@@ -110,7 +110,12 @@ __device__ __forceinline__ void salsa8_z(uint32_t &A, uint32_t &B,
     c ^= __builtin_rotateleft32(b + a, 9);
     d ^= __builtin_rotateleft32(c + b, 13);
     a ^= __builtin_rotateleft32(d + c, 18);
-    /* row round: QR(z0, rot1(z3), rot2(z2), rot3(z1)) */
+    /* row round: QR(z0, rot1(z3), rot2(z2), rot3(z1)).  The compiler
+     * fuses these permutes into the consuming xor/add as *_dpp forms; the
+     * VALU->DPP hazard s_nops that remain are issue-equivalent to unfused
+     * movs (measured: source order and dual-stream interleave both
+     * neutral — the kernel sits at the mixed read+write HBM ceiling,
+     * profiles/r01_kernel_profile.md). */
     uint32_t y1 = SWZ(d, QROT1);
     uint32_t y2 = SWZ(c, QROT2);
     uint32_t y3 = SWZ(b, QROT3);
@@ -144,9 +149,9 @@ __device__ __forceinline__ void salsa8_z2(uint32_t &A0, uint32_t &B0,
     d1 ^= __builtin_rotateleft32(c1 + b1, 13);
     a0 ^= __builtin_rotateleft32(d0 + c0, 18);
     a1 ^= __builtin_rotateleft32(d1 + c1, 18);
-    uint32_t y10 = SWZ(d0, QROT1), y11 = SWZ(d1, QROT1);
-    uint32_t y20 = SWZ(c0, QROT2), y21 = SWZ(c1, QROT2);
     uint32_t y30 = SWZ(b0, QROT3), y31 = SWZ(b1, QROT3);
+    uint32_t y20 = SWZ(c0, QROT2), y21 = SWZ(c1, QROT2);
+    uint32_t y10 = SWZ(d0, QROT1), y11 = SWZ(d1, QROT1);
     y10 ^= __builtin_rotateleft32(a0 + y30, 7);
     y11 ^= __builtin_rotateleft32(a1 + y31, 7);
     y20 ^= __builtin_rotateleft32(y10 + a0, 9);
