@@ -387,3 +387,36 @@ def test_verify_malformed_shapes():
             ver.verify(gsm_amd.PostProof(0, bad_indices, 0), meta)
         assert ei.value.code in (gsm_amd.api.Status.INVALID_ARGS,
                                  gsm_amd.api.Status.POW)
+
+
+def test_post_service_real_prover_roundtrip(tmp_path):
+    """Out-of-process placement: node server + real post-service child
+    proving over gRPC from a GPU-initialized data dir (the post-service
+    drop-in role)."""
+    import importlib
+    sup_mod = importlib.import_module("go-spacemesh_amd.supervisor")
+    d = str(tmp_path)
+    NU, LPU, N = 1, 1 << 12, 32
+    cfg, mgr = make_mgr(NU, LPU, N, data_dir=d, max_file_size=1 << 16)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    mgr.reset()
+
+    server = sup_mod.PostServiceServer()
+    sup = sup_mod.PostSupervisor(server.address, d, nonces=16)
+    sup.start()
+    try:
+        client = server.wait_for_client(timeout=30, poll_interval=0.2)
+        info = client.info()
+        assert info.node_id == NODE and info.num_units == NU
+        proof = client.proof(CHALLENGE, timeout=120)
+        # verify through the engine like the node's validator would
+        pcfg = gsm_amd.PostConfig(min_num_units=1, labels_per_unit=LPU,
+                                  k1=26, k2=37, k3=37)
+        ver = gsm_amd.PostVerifier(pcfg, scrypt_n=N)
+        meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, NU, LPU)
+        ver.verify(gsm_amd.PostProof(proof.nonce, proof.indices, proof.pow),
+                   meta)
+    finally:
+        sup.stop()
+        server.stop()
