@@ -461,11 +461,7 @@ def test_cfg1_checksum_regression():
     got = mgr.copy_labels(0, 1 << 20)
     mgr.reset()
     digest = hashlib.sha256(got).hexdigest()
-    expected = os.environ.get("POST_CFG1_CHECKSUM_OVERRIDE",
-                              "PRINT_ME")
-    if expected == "PRINT_ME":
-        print(f"\ncfg1-prefix sha256: {digest}")
-        known = "c2e1f3b2PLACEHOLDER"
-        if "PLACEHOLDER" in known:
-            pytest.skip(f"checksum recorded: {digest}")
+    # recorded on MI355X, round 1 (gpurun summary11)
+    expected = ("3cb88b0552f172abbf82a210509c3632"
+                "a1155df747b825b82334b67173219c4b")
     assert digest == expected
