@@ -465,3 +465,31 @@ def test_cfg1_checksum_regression():
     expected = ("3cb88b0552f172abbf82a210509c3632"
                 "a1155df747b825b82334b67173219c4b")
     assert digest == expected
+
+
+def test_session_cycle_no_leak():
+    """Create/run/free sessions and verify batches repeatedly; device free
+    memory must return to (near) its starting level — no leaked scratch."""
+    import torch
+    free0, _ = torch.cuda.mem_get_info(0)
+    for i in range(5):
+        cfg, mgr = make_mgr(1, 1 << 12, 128)
+        mgr.prepare_initializer()
+        mgr.start_session()
+        labels = mgr.copy_labels(0, 1 << 12)
+        proof = gsm_amd.api.prove_buffer(
+            labels, 1 << 12, NODE, ATX, CHALLENGE,
+            gsm_amd.PostConfig(min_num_units=1, labels_per_unit=1 << 12,
+                               k1=12, k2=8, pow_difficulty=POW_DIFF),
+            gsm_amd.ProveOpts(nonces=16))
+        mgr.reset()
+        ver = gsm_amd.PostVerifier(
+            gsm_amd.PostConfig(min_num_units=1, labels_per_unit=1 << 12,
+                               k1=12, k2=8, k3=8, pow_difficulty=POW_DIFF),
+            scrypt_n=128)
+        meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, 1, 1 << 12)
+        ver.verify(proof, meta)
+    free1, _ = torch.cuda.mem_get_info(0)
+    # the cached verify workspace and AES tables stay resident (bounded);
+    # everything else must be returned
+    assert free0 - free1 < (2 << 30), (free0, free1)
