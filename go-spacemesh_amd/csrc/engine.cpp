@@ -935,6 +935,115 @@ int post_prove_buffer(const uint8_t *labels, uint64_t num_labels,
   return prove_core(reader, num_labels, cfg, out);
 }
 
+int post_k2pow_search(const uint8_t challenge[32], uint32_t nonce_group,
+                      const uint8_t pow_difficulty[32], uint32_t pow_mode,
+                      uint32_t threads, uint64_t *out) {
+  if (pow_mode != POST_POW_MODE_BLAKE3) {
+    set_error("RandomX k2pow is not supported (use POST_POW_MODE_BLAKE3)");
+    return POST_ERR_UNSUPPORTED;
+  }
+  *out = poste::k2pow_search_blake3(challenge, nonce_group, pow_difficulty,
+                                    threads);
+  return POST_OK;
+}
+
+int post_prove_scan(const uint8_t *labels, uint64_t count,
+                    uint64_t index_base, uint64_t total_labels,
+                    const PostProveConfig *cfg, const uint64_t *group_pows,
+                    PostScanHitOut *hits, uint32_t cap, uint32_t *n_hits) {
+  if (!labels || !cfg || !group_pows || !hits || !n_hits)
+    return POST_ERR_INVALID_ARGS;
+  if (cfg->nonces == 0 || cfg->nonces % POSTE_NONCE_GROUP != 0) {
+    set_error("nonces must be a positive multiple of 16");
+    return POST_ERR_INVALID_ARGS;
+  }
+  int rc = require_gpu(cfg->provider_id);
+  if (rc != POST_OK) return rc;
+  DeviceTables tbl;
+  rc = get_aes_tables((int)cfg->provider_id, tbl);
+  if (rc != POST_OK) return rc;
+
+  const uint32_t n_ciphers = cfg->nonces / POSTE_NONCES_PER_AES;
+  std::vector<uint32_t> rk((size_t)n_ciphers * 44);
+  for (uint32_t c = 0; c < n_ciphers; c++) {
+    uint8_t key[16];
+    uint32_t grp = (c * POSTE_NONCES_PER_AES) / POSTE_NONCE_GROUP;
+    poste::prove_cipher_key(cfg->challenge, c, group_pows[grp], key);
+    poste::aes128_expand(key, rk.data() + (size_t)c * 44);
+  }
+  uint32_t *d_rk = nullptr;
+  uint8_t *d_labels = nullptr;
+  PostScanHit *d_hits = nullptr;
+  unsigned int *d_hit_count = nullptr;
+  auto cleanup = [&] {
+    if (d_rk) (void)hipFree(d_rk);
+    if (d_labels) (void)hipFree(d_labels);
+    if (d_hits) (void)hipFree(d_hits);
+    if (d_hit_count) (void)hipFree(d_hit_count);
+  };
+#undef HIP_TRY
+#define HIP_TRY(expr)                                                          \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      cleanup();                                                               \
+      return hip_fail(#expr, _e);                                              \
+    }                                                                          \
+  } while (0)
+  HIP_TRY(hipMalloc(&d_rk, rk.size() * 4));
+  HIP_TRY(hipMemcpy(d_rk, rk.data(), rk.size() * 4, hipMemcpyHostToDevice));
+  const uint64_t CHUNK = 1ull << 24;
+  HIP_TRY(hipMalloc(&d_labels, std::min(CHUNK, count) * 16));
+  HIP_TRY(hipMalloc(&d_hits, sizeof(PostScanHit) * cap));
+  HIP_TRY(hipMalloc(&d_hit_count, 4));
+  unsigned int zero = 0;
+  HIP_TRY(hipMemcpy(d_hit_count, &zero, 4, hipMemcpyHostToDevice));
+
+  ScanKernelArgs sa;
+  std::memset(&sa, 0, sizeof(sa));
+  sa.te = tbl.d_te;
+  sa.sbox = tbl.d_sbox;
+  sa.rk = d_rk;
+  sa.n_ciphers = n_ciphers;
+  sa.difficulty = poste::proving_difficulty(cfg->k1, total_labels);
+  sa.hits = d_hits;
+  sa.hit_count = d_hit_count;
+  sa.hit_cap = cap;
+  for (uint64_t base = 0; base < count; base += CHUNK) {
+    uint64_t cnt = std::min(CHUNK, count - base);
+    HIP_TRY(hipMemcpy(d_labels, labels + base * 16, cnt * 16,
+                      hipMemcpyHostToDevice));
+    sa.labels = (const uint4 *)d_labels;
+    sa.count = cnt;
+    sa.index_base = index_base + base;
+    uint32_t blocks = (uint32_t)std::min<uint64_t>(
+        (cnt + THREADS - 1) / THREADS, 8192);
+    HIP_TRY(poste_launch_scan_kernel(&sa, blocks, nullptr));
+    HIP_TRY(hipDeviceSynchronize());
+  }
+  unsigned int got = 0;
+  HIP_TRY(hipMemcpy(&got, d_hit_count, 4, hipMemcpyDeviceToHost));
+  if (got > cap) got = cap;
+  std::vector<PostScanHit> tmp(got);
+  if (got)
+    HIP_TRY(hipMemcpy(tmp.data(), d_hits, sizeof(PostScanHit) * got,
+                      hipMemcpyDeviceToHost));
+#undef HIP_TRY
+#define HIP_TRY(expr)                                                          \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) return hip_fail(#expr, _e);                          \
+  } while (0)
+  cleanup();
+  for (unsigned int i = 0; i < got; i++) {
+    hits[i].index = tmp[i].index;
+    hits[i].nonce = tmp[i].nonce;
+    hits[i].pad = 0;
+  }
+  *n_hits = got;
+  return POST_OK;
+}
+
 int post_prove(const char *data_dir, const PostProveConfig *cfg,
                PostProof *out) {
   if (!data_dir || !cfg || !out) return POST_ERR_INVALID_ARGS;

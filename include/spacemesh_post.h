@@ -178,12 +178,32 @@ int post_prove(const char *data_dir, const PostProveConfig *cfg,
                PostProof *out);
 
 /* Same, over a host buffer of 16-byte labels for index range
- * [0, num_labels) (bench/tests; BASELINE config 4 uses device-resident
- * labels via post_prove_device in engine-internal code). */
+ * [0, num_labels) (bench/tests). */
 int post_prove_buffer(const uint8_t *labels, uint64_t num_labels,
                       const uint8_t node_id[32],
                       const uint8_t commitment_atx_id[32],
                       const PostProveConfig *cfg, PostProof *out);
+
+/* Multi-GPU proving shard: scan labels [index_base,
+ * index_base + count) of a space of total_labels labels and return the
+ * raw passing (index, nonce) pairs; a coordinator merges shards and packs
+ * the winning nonce's indices (go-spacemesh_amd/proving.py).  group_pows
+ * must hold nonces/16 k2pow values for the challenge (computed once,
+ * shared by all shards). */
+typedef struct {
+  uint64_t index;
+  uint32_t nonce;
+  uint32_t pad;
+} PostScanHitOut;
+int post_prove_scan(const uint8_t *labels, uint64_t count,
+                    uint64_t index_base, uint64_t total_labels,
+                    const PostProveConfig *cfg, const uint64_t *group_pows,
+                    PostScanHitOut *hits, uint32_t cap, uint32_t *n_hits);
+
+/* k2pow helpers so a coordinator can compute/verify group pows once */
+int post_k2pow_search(const uint8_t challenge[32], uint32_t nonce_group,
+                      const uint8_t pow_difficulty[32], uint32_t pow_mode,
+                      uint32_t threads, uint64_t *out);
 
 /* ---------- verification ---------- */
 typedef struct {

@@ -493,3 +493,22 @@ def test_session_cycle_no_leak():
     # the cached verify workspace and AES tables stay resident (bounded);
     # everything else must be returned
     assert free0 - free1 < (2 << 30), (free0, free1)
+
+
+def test_sharded_prove_equals_whole(roundtrip):
+    """Two index-range scan shards merged on the host produce the exact
+    proof of the single-GPU prover (the 8-GPU proving axis)."""
+    import importlib
+    proving = importlib.import_module("go-spacemesh_amd.proving")
+    NU, LPU, N, labels, proof, _ = roundtrip
+    total = NU * LPU
+    cfg = gsm_amd.PostConfig(min_num_units=1, labels_per_unit=LPU,
+                             k1=12, k2=8, pow_difficulty=POW_DIFF)
+    cut = 700 * 16  # ragged split
+    merged = proving.prove_sharded(
+        [(labels[:cut], 0), (labels[cut:], 700)], total, CHALLENGE, cfg,
+        nonces=16)
+    assert merged is not None
+    assert merged.nonce == proof.nonce
+    assert merged.pow == proof.pow
+    assert merged.indices == proof.indices
