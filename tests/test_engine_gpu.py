@@ -512,3 +512,42 @@ def test_sharded_prove_equals_whole(roundtrip):
     assert merged.nonce == proof.nonce
     assert merged.pow == proof.pow
     assert merged.indices == proof.indices
+
+
+def test_romix2_env_path_parity(tmp_path):
+    """The POST_ROMIX2=1 dual-stream kernel path stays bit-exact (guards
+    the alternate path against rot).  Runs in a subprocess because the
+    launcher caches the env choice per process."""
+    import subprocess
+    import sys
+    script = r"""
+import sys
+sys.path.insert(0, %r)
+sys.path.insert(0, %r)
+import gsm_amd
+from oracle import Oracle
+NODE, ATX = bytes([0xA5])*32, bytes([0x5A])*32
+POW = bytes([0x0F]) + bytes([0xFF])*31
+cfg = gsm_amd.PostConfig(min_num_units=1, labels_per_unit=1500, k1=12,
+                         k2=8, k3=4, pow_difficulty=POW)
+opts = gsm_amd.PostSetupOpts(num_units=1, scrypt_n=128,
+                             scratch_bytes=8 << 30)
+mgr = gsm_amd.PostSetupManager(NODE, ATX, cfg, opts)
+mgr.prepare_initializer()
+mgr.start_session()
+got = mgr.copy_labels(0, 1500)
+o = Oracle()
+want, best = o.init_range(o.commitment(NODE, ATX), 0, 1500, 128)
+assert got == want, "labels mismatch under POST_ROMIX2"
+n = mgr.vrf_nonce()
+assert n is not None and n[0] == best.index
+mgr.reset()
+print("romix2 parity OK")
+"""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, POST_ROMIX2="1")
+    r = subprocess.run(
+        [sys.executable, "-c", script % (repo, os.path.join(repo, "oracle"))],
+        capture_output=True, text=True, timeout=240, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "romix2 parity OK" in r.stdout
