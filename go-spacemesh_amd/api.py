@@ -23,6 +23,7 @@ from ctypes import (POINTER, byref, c_char_p, c_int, c_int32, c_size_t,
 from typing import Optional
 
 from . import events as _ev
+from . import metrics as _metrics
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
 _LIB_PATH = os.environ.get("POST_ENGINE_LIB",
@@ -337,6 +338,7 @@ class PostSetupManager:
             raise EngineError(Status.ERR, "post session not prepared")
         self.state = self.IN_PROGRESS
         _ev.bus().emit(_ev.InitStart(self.node_id, self.commitment_atx_id))
+        _metrics.init_size.labels(step="start").set(self.opts.num_units)
         rc = self.engine.lib.post_init_run(self._session)
         if rc == Status.CANCELLED:
             self.state = self.STOPPED
@@ -347,6 +349,7 @@ class PostSetupManager:
             _ev.bus().emit(_ev.InitFailure(self.node_id, err))
             raise EngineError(rc, err)
         self.state = self.COMPLETE
+        _metrics.init_size.labels(step="complete").set(self.opts.num_units)
         _ev.bus().emit(_ev.InitComplete(self.node_id))
 
     def step(self, max_labels: int) -> tuple:
@@ -400,11 +403,16 @@ class PostSetupManager:
 def prove_buffer(labels: bytes, num_labels: int, node_id: bytes,
                  atx_id: bytes, challenge: bytes, cfg: PostConfig,
                  opts: ProveOpts) -> PostProof:
+    import time as _time
     _ev.bus().emit(_ev.PostStart(node_id, challenge))
+    t0 = _time.monotonic()
     try:
         return _prove_buffer(labels, num_labels, node_id, atx_id, challenge,
                              cfg, opts)
     finally:
+        dt = _time.monotonic() - t0
+        _metrics.post_seconds.set(dt)          # smh_post_seconds
+        _metrics.post_duration.set(dt * 1e9)   # activation_post_duration
         _ev.bus().emit(_ev.PostComplete(node_id))
 
 

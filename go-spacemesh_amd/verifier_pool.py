@@ -22,6 +22,7 @@ import threading
 from typing import Iterable, Optional
 
 from . import events as ev
+from . import metrics as _metrics
 
 
 class _Job:
@@ -129,8 +130,16 @@ class OffloadingVerifier:
                 raise RuntimeError("verifier closed")
         own = prioritized or bytes(meta.node_id) in self._prioritized
         job = _Job(proof, meta, opts)
-        self._q.put((0 if own else 1, next(self._seq), job))
-        job.done.wait()
+        _metrics.post_verification_queue.inc()  # post_verifier.go:319
+        import time as _time
+        t0 = _time.monotonic()
+        try:
+            self._q.put((0 if own else 1, next(self._seq), job))
+            job.done.wait()
+        finally:
+            _metrics.post_verification_queue.dec()
+            _metrics.post_verification_latency.observe(
+                _time.monotonic() - t0)  # validation.go:216-220
         if job.error is not None:
             raise job.error
 
