@@ -551,3 +551,51 @@ print("romix2 parity OK")
         capture_output=True, text=True, timeout=240, env=env)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "romix2 parity OK" in r.stdout
+
+
+def test_randomized_config_fuzz_parity():
+    """Randomized-config round-trip fuzz (the reference runs fuzzing over
+    its codecs/state, Makefile:195; here: random scrypt-N / space sizes /
+    K-params, engine vs oracle end to end, fixed seed)."""
+    import random
+    rng = random.Random(0xF00D)
+    o = Oracle()
+    for trial in range(5):
+        n = 1 << rng.randrange(1, 10)         # scrypt N in 2..512
+        total = rng.randrange(200, 2500)
+        k2 = rng.randrange(4, 16)
+        k1 = k2 + rng.randrange(0, 8)
+        nonces = 16 * rng.randrange(1, 3)
+        node = bytes(rng.randrange(256) for _ in range(32))
+        atx = bytes(rng.randrange(256) for _ in range(32))
+        challenge = bytes(rng.randrange(256) for _ in range(32))
+        cfg = gsm_amd.PostConfig(min_num_units=1, labels_per_unit=total,
+                                 k1=k1, k2=k2, k3=k2,
+                                 pow_difficulty=POW_DIFF)
+        opts = gsm_amd.PostSetupOpts(num_units=1, scrypt_n=n,
+                                     scratch_bytes=8 << 30)
+        mgr = gsm_amd.PostSetupManager(node, atx, cfg, opts)
+        mgr.prepare_initializer()
+        mgr.start_session()
+        got = mgr.copy_labels(0, total)
+        nonce = mgr.vrf_nonce()
+        mgr.reset()
+        commit = o.commitment(node, atx)
+        want, best = o.init_range(commit, 0, total, n)
+        assert got == want, (trial, n, total)
+        assert nonce is not None and nonce[0] == best.index, (trial,)
+        try:
+            proof = gsm_amd.api.prove_buffer(
+                got, total, node, atx, challenge, cfg,
+                gsm_amd.ProveOpts(nonces=nonces))
+        except gsm_amd.EngineError as e:
+            assert e.code == gsm_amd.api.Status.NO_NONCE
+            with pytest.raises(ValueError):
+                o.prove(got, total, challenge, k1, k2, nonces, POW_DIFF)
+            continue
+        op = o.prove(got, total, challenge, k1, k2, nonces, POW_DIFF)
+        assert (proof.nonce, proof.pow) == (op.nonce, op.pow), (trial,)
+        assert proof.indices == bytes(op.indices[:op.indices_len])
+        ver = gsm_amd.PostVerifier(cfg, scrypt_n=n)
+        meta = gsm_amd.PostProofMetadata(node, atx, challenge, 1, total)
+        ver.verify(proof, meta)
