@@ -96,6 +96,10 @@ def main():
                 {"N": n, "index": idx, "full": full.hex()})
     out["labels_openssl"] = label_vectors
 
+    # NOTE: the "protocol_frozen" section (k2pow/cipher-key/subset/
+    # difficulty/proof fixtures freezing the RESTATED post-rs semantics) is
+    # appended by the snippet recorded in the r01 history; regenerating this
+    # file drops it — re-add before committing.
     with open(os.path.join(HERE, "golden.json"), "w") as f:
         json.dump(out, f, indent=1)
     print("wrote", os.path.join(HERE, "golden.json"))

@@ -599,3 +599,32 @@ def test_randomized_config_fuzz_parity():
         ver = gsm_amd.PostVerifier(cfg, scrypt_n=n)
         meta = gsm_amd.PostProofMetadata(node, atx, challenge, 1, total)
         ver.verify(proof, meta)
+
+
+def test_engine_reproduces_frozen_proof_fixture():
+    """The engine end-to-end reproduces the committed protocol fixture
+    (tests/golden 'protocol_frozen'): init checksum, VRF nonce, proof."""
+    import hashlib
+    fx = GOLDEN["protocol_frozen"]["proof_fixture"]
+    node = bytes.fromhex(fx["node_id"])
+    atx = bytes.fromhex(fx["atx_id"])
+    cfg = gsm_amd.PostConfig(
+        min_num_units=1, labels_per_unit=fx["num_labels"], k1=fx["k1"],
+        k2=fx["k2"], k3=fx["k2"],
+        pow_difficulty=bytes.fromhex(fx["pow_difficulty"]))
+    opts = gsm_amd.PostSetupOpts(num_units=1, scrypt_n=fx["scrypt_n"],
+                                 scratch_bytes=4 << 30)
+    mgr = gsm_amd.PostSetupManager(node, atx, cfg, opts)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    labels = mgr.copy_labels(0, fx["num_labels"])
+    assert hashlib.sha256(labels).hexdigest() == fx["labels_sha256"]
+    nonce = mgr.vrf_nonce()
+    assert nonce is not None and nonce[0] == fx["vrf_nonce_index"]
+    mgr.reset()
+    proof = gsm_amd.api.prove_buffer(
+        labels, fx["num_labels"], node, atx, bytes.fromhex(fx["challenge"]),
+        cfg, gsm_amd.ProveOpts(nonces=fx["nonces"]))
+    assert proof.nonce == fx["proof_nonce"]
+    assert proof.pow == fx["proof_pow"]
+    assert proof.indices.hex() == fx["proof_indices"]
