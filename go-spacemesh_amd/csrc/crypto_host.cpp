@@ -438,6 +438,7 @@ uint64_t k2pow_search_blake3(const uint8_t challenge[32], uint32_t nonce_group,
 }
 
 uint64_t proving_difficulty(uint32_t k1, uint64_t num_labels) {
+  if (num_labels == 0) return 0; /* malformed metadata: nothing passes */
   unsigned __int128 d = ((unsigned __int128)k1 << 64) / num_labels;
   return d > (unsigned __int128)UINT64_MAX ? UINT64_MAX : (uint64_t)d;
 }

@@ -1133,6 +1133,10 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
     const PostProof &pr = proofs[p];
     const PostProofMetadata &me = metas[p];
     uint64_t num_labels = (uint64_t)me.num_units * me.labels_per_unit;
+    if (num_labels == 0) { /* malformed metadata */
+      statuses[p] = POST_ERR_INVALID_ARGS;
+      continue;
+    }
     uint32_t bpi = poste::bits_per_index(num_labels);
     if (pr.num_indices != cfg->k2 ||
         pr.indices_len != ((uint64_t)cfg->k2 * bpi + 7) / 8) {

@@ -94,6 +94,7 @@ void oracle_vrf_difficulty(uint64_t num_labels, uint8_t out[32]) {
 
 uint64_t oracle_proving_difficulty(uint32_t k1, uint64_t num_labels) {
   /* floor(k1 * 2^64 / num_labels) (RESTATED) */
+  if (num_labels == 0) return 0; /* malformed metadata: nothing passes */
   unsigned __int128 x = (unsigned __int128)k1 << 64;
   unsigned __int128 d = x / num_labels;
   if (d > (unsigned __int128)UINT64_MAX) return UINT64_MAX;
@@ -274,6 +275,7 @@ int oracle_verify(const OracleProof *proof, const OracleProofMetadata *meta,
                   int32_t selected_index, const uint8_t pow_difficulty[32],
                   uint32_t *invalid_index) {
   uint64_t num_labels = (uint64_t)meta->num_units * meta->labels_per_unit;
+  if (num_labels == 0) return ORACLE_VERIFY_ERR_MALFORMED;
   if (proof->num_indices != k2) return ORACLE_VERIFY_ERR_MALFORMED;
   uint32_t bpi = oracle_bits_per_index(num_labels);
   if (proof->indices_len != ((uint64_t)k2 * bpi + 7) / 8)

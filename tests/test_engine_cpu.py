@@ -106,3 +106,15 @@ def test_randomx_mode_unsupported_is_explicit():
     # but the error must be one of the two explicit codes
     assert ei.value.code in (gsm_amd.api.Status.UNSUPPORTED,
                              gsm_amd.api.Status.NO_GPU)
+
+
+def test_zero_labels_metadata_rejected(oracle):
+    """num_units*labels_per_unit == 0 must be rejected, not divide by zero
+    (SIGFPE) in the difficulty computation."""
+    from oracle import Proof, make_meta
+    assert oracle.lib.oracle_proving_difficulty(26, 0) == 0
+    meta = make_meta(bytes(32), bytes(32), bytes(32), 0, 64)
+    p = Proof()
+    p.num_indices = 8
+    rc, _ = oracle.verify(p, meta, 2, 12, 8, 8, None, -1, bytes(32))
+    assert rc == 3  # malformed
