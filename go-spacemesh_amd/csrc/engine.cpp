@@ -874,7 +874,12 @@ static int prove_core(const LabelReader &read_labels, uint64_t num_labels,
   HIP_TRY(hipStreamSynchronize(stream));
   unsigned int n_hits = 0;
   HIP_TRY(hipMemcpy(&n_hits, d_hit_count, 4, hipMemcpyDeviceToHost));
-  if (n_hits > HIT_CAP) n_hits = HIT_CAP;
+  if (n_hits > HIT_CAP) {
+    cleanup();
+    set_error("scan hit buffer overflow (pathological K1/num_labels "
+              "configuration)");
+    return POST_ERR;
+  }
   std::vector<PostScanHit> all_hits(n_hits);
   if (n_hits)
     HIP_TRY(hipMemcpy(all_hits.data(), d_hits, sizeof(PostScanHit) * n_hits,
@@ -1023,7 +1028,11 @@ int post_prove_scan(const uint8_t *labels, uint64_t count,
   }
   unsigned int got = 0;
   HIP_TRY(hipMemcpy(&got, d_hit_count, 4, hipMemcpyDeviceToHost));
-  if (got > cap) got = cap;
+  if (got > cap) {
+    cleanup();
+    set_error("scan hit buffer overflow (raise cap)");
+    return POST_ERR;
+  }
   std::vector<PostScanHit> tmp(got);
   if (got)
     HIP_TRY(hipMemcpy(tmp.data(), d_hits, sizeof(PostScanHit) * got,
@@ -1132,6 +1141,16 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
     statuses[p] = POST_OK;
     const PostProof &pr = proofs[p];
     const PostProofMetadata &me = metas[p];
+    /* the commitment table is indexed by ABSOLUTE proof number from the
+     * kernel tasks, so every proof gets a slot — including ones rejected
+     * below */
+    {
+      uint8_t cm[32];
+      poste::commitment(me.node_id, me.commitment_atx_id, cm);
+      uint32_t cw[8];
+      std::memcpy(cw, cm, 32);
+      for (int k = 0; k < 8; k++) commit_words.push_back(cw[k]);
+    }
     uint64_t num_labels = (uint64_t)me.num_units * me.labels_per_unit;
     if (num_labels == 0) { /* malformed metadata */
       statuses[p] = POST_ERR_INVALID_ARGS;
@@ -1152,11 +1171,6 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
     proof_indices[p].resize(cfg->k2);
     poste::unpack_indices(pr.indices, cfg->k2, bpi,
                           proof_indices[p].data());
-    uint8_t cm[32];
-    poste::commitment(me.node_id, me.commitment_atx_id, cm);
-    uint32_t cw[8];
-    std::memcpy(cw, cm, 32);
-    for (int k = 0; k < 8; k++) commit_words.push_back(cw[k]);
 
     std::vector<uint32_t> positions;
     if (cfg->selected_index >= 0) {

@@ -628,3 +628,20 @@ def test_engine_reproduces_frozen_proof_fixture():
     assert proof.nonce == fx["proof_nonce"]
     assert proof.pow == fx["proof_pow"]
     assert proof.indices.hex() == fx["proof_indices"]
+
+
+def test_verify_batch_commitments_survive_early_rejects(roundtrip):
+    """Regression: a proof rejected at the pow/shape stage must not skew the
+    commitment table of the surviving proofs (absolute-index table)."""
+    NU, LPU, N, labels, proof, _ = roundtrip
+    meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, NU, LPU)
+    bad_pow = gsm_amd.PostProof(proof.nonce, proof.indices, proof.pow + 1)
+    bad_shape = gsm_amd.PostProof(proof.nonce, proof.indices[:-2], proof.pow)
+    cfg = gsm_amd.PostConfig(k1=12, k2=8, k3=8, pow_difficulty=POW_DIFF)
+    ver = gsm_amd.PostVerifier(cfg, scrypt_n=N)
+    res = ver.verify_batch([bad_pow, proof, bad_shape, proof],
+                           [meta, meta, meta, meta])
+    assert res[0][0] == gsm_amd.api.Status.POW
+    assert res[1][0] == gsm_amd.api.Status.OK
+    assert res[2][0] == gsm_amd.api.Status.INVALID_ARGS
+    assert res[3][0] == gsm_amd.api.Status.OK
