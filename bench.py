@@ -86,12 +86,16 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    # RCCL in production; POST_BENCH_BACKEND=gloo lets the world>1 rank
+    # logic run with several ranks sharing one GPU (functional testing)
+    backend = os.environ.get("POST_BENCH_BACKEND", "nccl")
+    device_id = local_rank % torch.cuda.device_count()
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
+        torch.cuda.set_device(device_id)
+        dist.init_process_group(backend)
 
     import gsm_amd
     from importlib import import_module
@@ -102,8 +106,9 @@ def main():
 
     cfg = gsm_amd.PostConfig()  # mainnet params
     opts = gsm_amd.PostSetupOpts(
-        num_units=NUM_UNITS, scrypt_n=SCRYPT_N, provider_id=local_rank,
-        index_start=start, index_end=end, data_dir=None)
+        num_units=NUM_UNITS, scrypt_n=SCRYPT_N, provider_id=device_id,
+        index_start=start, index_end=end, data_dir=None,
+        scratch_bytes=int(os.environ.get("POST_BENCH_SCRATCH", "0")))
     mgr = gsm_amd.PostSetupManager(bytes([0xA5]) * 32, bytes([0x5A]) * 32,
                                    cfg, opts)
     mgr.prepare_initializer()
@@ -128,7 +133,8 @@ def main():
 
     # whole-job time = MAX over ranks
     if dist:
-        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dev = "cuda" if backend == "nccl" else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
