@@ -17,6 +17,13 @@ HDRS_ENGINE := $(CSRC)/post_common.h $(CSRC)/kernel_args.h \
 $(ENGINE): $(SRCS_ENGINE) $(HDRS_ENGINE)
 	$(HIPCC) $(HIPFLAGS) -shared $(SRCS_ENGINE) -o $@
 
+# A/B variant: non-temporal scratch access (POST_ENGINE_LIB selects)
+go-spacemesh_amd/libpost_hip_nt.so: $(SRCS_ENGINE) $(HDRS_ENGINE)
+	$(HIPCC) $(HIPFLAGS) -DPOSTE_NT=1 -shared $(SRCS_ENGINE) -o $@
+
+nt: go-spacemesh_amd/libpost_hip_nt.so
+.PHONY: nt
+
 oracle:
 	$(MAKE) -C oracle
 

@@ -27,6 +27,26 @@
 
 #define POSTE_THREADS 256
 
+/* ROMix scratch V is written once and read once ~N iterations later — far
+ * beyond any cache. -DPOSTE_NT=1 builds with non-temporal loads/stores on
+ * V to keep it out of L2 (A/B via POST_ENGINE_LIB). */
+#if defined(POSTE_NT) && POSTE_NT
+typedef uint32_t poste_v4u __attribute__((ext_vector_type(4)));
+#define VSTORE(p, v)                                                           \
+  do {                                                                         \
+    uint4 _t = (v);                                                            \
+    __builtin_nontemporal_store(*(poste_v4u *)&_t, (poste_v4u *)(p));          \
+  } while (0)
+#define VLOAD(p)                                                               \
+  ({                                                                           \
+    poste_v4u _r = __builtin_nontemporal_load((const poste_v4u *)(p));         \
+    *(uint4 *)&_r;                                                             \
+  })
+#else
+#define VSTORE(p, v) (*(p) = (v))
+#define VLOAD(p) (*(p))
+#endif
+
 /* ------------------------- SHA-256 (device) ------------------------- */
 __constant__ uint32_t c_sha_k[64] = {
     0x428a2f98, 0x71374491, 0xb5c0fbcf, 0xe9b5dba5, 0x3956c25b, 0x59f111f1,
@@ -358,8 +378,8 @@ post_label_romix_kernel(LabelKernelArgs a) {
       for (uint32_t j = 0; j < n; j++) {
         if ((j & gmask) == 0) {
           uint4 *p = V + base;
-          p[0] = make_uint4(Z0[0], Z0[1], Z0[2], Z0[3]);
-          p[4] = make_uint4(Z1[0], Z1[1], Z1[2], Z1[3]);
+          VSTORE(p, make_uint4(Z0[0], Z0[1], Z0[2], Z0[3]));
+          VSTORE(p + 4, make_uint4(Z1[0], Z1[1], Z1[2], Z1[3]));
           base += stride;
         }
         blockmix_z(Z0, Z1);
@@ -376,7 +396,7 @@ post_label_romix_kernel(LabelKernelArgs a) {
                group) * 8ull + sub;
       uint32_t Y0[4], Y1[4];
       {
-        uint4 v0 = p[0], v1 = p[4];
+        uint4 v0 = VLOAD(p), v1 = VLOAD(p + 4);
         Y0[0] = v0.x; Y0[1] = v0.y; Y0[2] = v0.z; Y0[3] = v0.w;
         Y1[0] = v1.x; Y1[1] = v1.y; Y1[2] = v1.z; Y1[3] = v1.w;
       }
