@@ -656,69 +656,100 @@ post_scan_kernel(ScanKernelArgs a) {
     sSbox[i] = a.sbox[i];
   __syncthreads();
 
+  /* Four independent labels per lane (ILP): a single AES chain leaves the
+   * wave ~76% latency-stalled (profiles: SQ_ACTIVE_INST 10%, LDS 14%);
+   * interleaving L chains fills the LDS-latency shadows. */
   const unsigned long long stride =
       (unsigned long long)gridDim.x * blockDim.x;
-  for (unsigned long long t =
+  constexpr int L = 4;
+  const unsigned long long span = stride * L;
+  for (unsigned long long t0 =
            (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
-       t < a.count; t += stride) {
-    uint4 lraw = a.labels[t];
-    const uint32_t p0 = __builtin_bswap32(lraw.x),
-                   p1 = __builtin_bswap32(lraw.y),
-                   p2 = __builtin_bswap32(lraw.z),
-                   p3 = __builtin_bswap32(lraw.w);
+       t0 < a.count; t0 += span) {
+    uint32_t p[L][4];
+    unsigned long long ts[L];
+#pragma unroll
+    for (int u = 0; u < L; u++) {
+      unsigned long long t = t0 + (unsigned long long)u * stride;
+      ts[u] = t < a.count ? t : t0; /* clamp: duplicates are harmless
+                                       (hits deduped by nonce list merge
+                                       never occur: same index+nonce hits
+                                       only appended for u==0's slot) */
+      uint4 lraw = a.labels[ts[u]];
+      p[u][0] = __builtin_bswap32(lraw.x);
+      p[u][1] = __builtin_bswap32(lraw.y);
+      p[u][2] = __builtin_bswap32(lraw.z);
+      p[u][3] = __builtin_bswap32(lraw.w);
+    }
+    const int live = (int)((a.count - t0 + stride - 1) / stride) < L
+                         ? (int)((a.count - t0 + stride - 1) / stride)
+                         : L;
     for (uint32_t c = 0; c < a.n_ciphers; c++) {
       const uint32_t *rk = a.rk + c * 44;
-      uint32_t w0 = p0 ^ rk[0], w1 = p1 ^ rk[1], w2 = p2 ^ rk[2],
-               w3 = p3 ^ rk[3];
+      uint32_t w[L][4];
+#pragma unroll
+      for (int u = 0; u < L; u++)
+#pragma unroll
+        for (int k = 0; k < 4; k++) w[u][k] = p[u][k] ^ rk[k];
 #pragma unroll
       for (int r = 1; r < 10; r++) {
-        uint32_t n0 = sTe[w0 >> 24] ^ sTe[256 + ((w1 >> 16) & 0xff)] ^
-                      sTe[512 + ((w2 >> 8) & 0xff)] ^ sTe[768 + (w3 & 0xff)] ^
-                      rk[4 * r];
-        uint32_t n1 = sTe[w1 >> 24] ^ sTe[256 + ((w2 >> 16) & 0xff)] ^
-                      sTe[512 + ((w3 >> 8) & 0xff)] ^ sTe[768 + (w0 & 0xff)] ^
-                      rk[4 * r + 1];
-        uint32_t n2 = sTe[w2 >> 24] ^ sTe[256 + ((w3 >> 16) & 0xff)] ^
-                      sTe[512 + ((w0 >> 8) & 0xff)] ^ sTe[768 + (w1 & 0xff)] ^
-                      rk[4 * r + 2];
-        uint32_t n3 = sTe[w3 >> 24] ^ sTe[256 + ((w0 >> 16) & 0xff)] ^
-                      sTe[512 + ((w1 >> 8) & 0xff)] ^ sTe[768 + (w2 & 0xff)] ^
-                      rk[4 * r + 3];
-        w0 = n0; w1 = n1; w2 = n2; w3 = n3;
-      }
-      uint32_t f0 = (((uint32_t)sSbox[w0 >> 24] << 24) |
-                     ((uint32_t)sSbox[(w1 >> 16) & 0xff] << 16) |
-                     ((uint32_t)sSbox[(w2 >> 8) & 0xff] << 8) |
-                     sSbox[w3 & 0xff]) ^ rk[40];
-      uint32_t f1 = (((uint32_t)sSbox[w1 >> 24] << 24) |
-                     ((uint32_t)sSbox[(w2 >> 16) & 0xff] << 16) |
-                     ((uint32_t)sSbox[(w3 >> 8) & 0xff] << 8) |
-                     sSbox[w0 & 0xff]) ^ rk[41];
-      uint32_t f2 = (((uint32_t)sSbox[w2 >> 24] << 24) |
-                     ((uint32_t)sSbox[(w3 >> 16) & 0xff] << 16) |
-                     ((uint32_t)sSbox[(w0 >> 8) & 0xff] << 8) |
-                     sSbox[w1 & 0xff]) ^ rk[42];
-      uint32_t f3 = (((uint32_t)sSbox[w3 >> 24] << 24) |
-                     ((uint32_t)sSbox[(w0 >> 16) & 0xff] << 16) |
-                     ((uint32_t)sSbox[(w1 >> 8) & 0xff] << 8) |
-                     sSbox[w2 & 0xff]) ^ rk[43];
-      /* per-nonce values: LE u64 of output bytes [0..8) and [8..16) */
-      unsigned long long v0 =
-          __builtin_bswap64(((unsigned long long)f0 << 32) | f1);
-      unsigned long long v1 =
-          __builtin_bswap64(((unsigned long long)f2 << 32) | f3);
-      if (v0 < a.difficulty) {
-        unsigned int s = atomicAdd(a.hit_count, 1u);
-        if (s < a.hit_cap) {
-          a.hits[s].index = a.index_base + t;
-          a.hits[s].nonce = c * POSTE_NONCES_PER_AES;
+#pragma unroll
+        for (int u = 0; u < L; u++) {
+          uint32_t n0 = sTe[w[u][0] >> 24] ^
+                        sTe[256 + ((w[u][1] >> 16) & 0xff)] ^
+                        sTe[512 + ((w[u][2] >> 8) & 0xff)] ^
+                        sTe[768 + (w[u][3] & 0xff)] ^ rk[4 * r];
+          uint32_t n1 = sTe[w[u][1] >> 24] ^
+                        sTe[256 + ((w[u][2] >> 16) & 0xff)] ^
+                        sTe[512 + ((w[u][3] >> 8) & 0xff)] ^
+                        sTe[768 + (w[u][0] & 0xff)] ^ rk[4 * r + 1];
+          uint32_t n2 = sTe[w[u][2] >> 24] ^
+                        sTe[256 + ((w[u][3] >> 16) & 0xff)] ^
+                        sTe[512 + ((w[u][0] >> 8) & 0xff)] ^
+                        sTe[768 + (w[u][1] & 0xff)] ^ rk[4 * r + 2];
+          uint32_t n3 = sTe[w[u][3] >> 24] ^
+                        sTe[256 + ((w[u][0] >> 16) & 0xff)] ^
+                        sTe[512 + ((w[u][1] >> 8) & 0xff)] ^
+                        sTe[768 + (w[u][2] & 0xff)] ^ rk[4 * r + 3];
+          w[u][0] = n0; w[u][1] = n1; w[u][2] = n2; w[u][3] = n3;
         }
       }
-      if (v1 < a.difficulty) {
-        unsigned int s = atomicAdd(a.hit_count, 1u);
-        if (s < a.hit_cap) {
-          a.hits[s].index = a.index_base + t;
-          a.hits[s].nonce = c * POSTE_NONCES_PER_AES + 1;
+#pragma unroll
+      for (int u = 0; u < L; u++) {
+        if (u >= live) break;
+        uint32_t f0 = (((uint32_t)sSbox[w[u][0] >> 24] << 24) |
+                       ((uint32_t)sSbox[(w[u][1] >> 16) & 0xff] << 16) |
+                       ((uint32_t)sSbox[(w[u][2] >> 8) & 0xff] << 8) |
+                       sSbox[w[u][3] & 0xff]) ^ rk[40];
+        uint32_t f1 = (((uint32_t)sSbox[w[u][1] >> 24] << 24) |
+                       ((uint32_t)sSbox[(w[u][2] >> 16) & 0xff] << 16) |
+                       ((uint32_t)sSbox[(w[u][3] >> 8) & 0xff] << 8) |
+                       sSbox[w[u][0] & 0xff]) ^ rk[41];
+        uint32_t f2 = (((uint32_t)sSbox[w[u][2] >> 24] << 24) |
+                       ((uint32_t)sSbox[(w[u][3] >> 16) & 0xff] << 16) |
+                       ((uint32_t)sSbox[(w[u][0] >> 8) & 0xff] << 8) |
+                       sSbox[w[u][1] & 0xff]) ^ rk[42];
+        uint32_t f3 = (((uint32_t)sSbox[w[u][3] >> 24] << 24) |
+                       ((uint32_t)sSbox[(w[u][0] >> 16) & 0xff] << 16) |
+                       ((uint32_t)sSbox[(w[u][1] >> 8) & 0xff] << 8) |
+                       sSbox[w[u][2] & 0xff]) ^ rk[43];
+        unsigned long long v0 =
+            __builtin_bswap64(((unsigned long long)f0 << 32) | f1);
+        unsigned long long v1 =
+            __builtin_bswap64(((unsigned long long)f2 << 32) | f3);
+        if (v0 < a.difficulty) {
+          unsigned int s = atomicAdd(a.hit_count, 1u);
+          if (s < a.hit_cap) {
+            a.hits[s].index = a.index_base + ts[u];
+            a.hits[s].nonce = c * POSTE_NONCES_PER_AES;
+          }
+        }
+        if (v1 < a.difficulty) {
+          unsigned int s = atomicAdd(a.hit_count, 1u);
+          if (s < a.hit_cap) {
+            a.hits[s].index = a.index_base + ts[u];
+            a.hits[s].nonce = c * POSTE_NONCES_PER_AES + 1;
+          }
         }
       }
     }
