@@ -24,6 +24,12 @@ go-spacemesh_amd/libpost_hip_nt.so: $(SRCS_ENGINE) $(HDRS_ENGINE)
 nt: go-spacemesh_amd/libpost_hip_nt.so
 .PHONY: nt
 
+# A/B variant: scan ILP depth 4
+go-spacemesh_amd/libpost_hip_scan4.so: $(SRCS_ENGINE) $(HDRS_ENGINE)
+	$(HIPCC) $(HIPFLAGS) -DPOSTE_SCAN_ILP=4 -shared $(SRCS_ENGINE) -o $@
+scan4: go-spacemesh_amd/libpost_hip_scan4.so
+.PHONY: scan4
+
 oracle:
 	$(MAKE) -C oracle
 

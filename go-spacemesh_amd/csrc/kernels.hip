@@ -661,7 +661,10 @@ post_scan_kernel(ScanKernelArgs a) {
    * interleaving L chains fills the LDS-latency shadows. */
   const unsigned long long stride =
       (unsigned long long)gridDim.x * blockDim.x;
-  constexpr int L = 4;
+#ifndef POSTE_SCAN_ILP
+#define POSTE_SCAN_ILP 2 /* 2 chains @ 8 waves/SIMD measured best */
+#endif
+  constexpr int L = POSTE_SCAN_ILP;
   const unsigned long long span = stride * L;
   for (unsigned long long t0 =
            (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
