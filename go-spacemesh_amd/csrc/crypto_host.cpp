@@ -412,10 +412,20 @@ int k2pow_verify_blake3(const uint8_t challenge[32], uint32_t nonce_group,
 uint64_t k2pow_search_blake3(const uint8_t challenge[32], uint32_t nonce_group,
                              const uint8_t pow_difficulty[32],
                              uint32_t threads) {
+  /* Realistic difficulties (mainnet ~2^-12 per try) are found within a few
+   * thousand sequential hashes (~ms); threads are only worth their spawn
+   * cost for much harder settings, so parallelize after a sequential
+   * probe window. */
+  const uint64_t SEQ_WINDOW = 1 << 17;
+  for (uint64_t p = 0; p < SEQ_WINDOW; p++) {
+    if (k2pow_verify_blake3(challenge, nonce_group, p, pow_difficulty) == 0)
+      return p;
+  }
   if (threads == 0) threads = std::thread::hardware_concurrency();
   if (threads == 0) threads = 1;
-  const uint64_t CHUNK = 16384;
-  for (uint64_t base = 0;; base += CHUNK * threads) {
+  if (threads > 64) threads = 64;
+  const uint64_t CHUNK = 1 << 18;
+  for (uint64_t base = SEQ_WINDOW;; base += CHUNK * threads) {
     std::atomic<uint64_t> best{UINT64_MAX};
     std::vector<std::thread> ts;
     for (uint32_t t = 0; t < threads; t++) {
