@@ -45,8 +45,17 @@ def cmd_benchmark(args):
 
 
 def _mk_cfg(args):
-    cfg = gsm_amd.PostConfig(labels_per_unit=args.labels_per_unit,
-                             min_num_units=1)
+    if getattr(args, "preset", None):
+        import importlib
+        presets = importlib.import_module("go-spacemesh_amd.presets")
+        cfg, popts = presets.get(args.preset)
+        args.labels_per_unit = cfg.labels_per_unit
+        args.scrypt_n = popts.scrypt_n
+        if args.num_units is None:
+            args.num_units = popts.num_units
+    else:
+        cfg = gsm_amd.PostConfig(labels_per_unit=args.labels_per_unit,
+                                 min_num_units=1)
     if getattr(args, "pow_difficulty", None):
         cfg.pow_difficulty = bytes.fromhex(args.pow_difficulty)
     return cfg
@@ -63,6 +72,9 @@ def cmd_init(args):
         r, w = (int(x) for x in args.shard.split("/"))
         total = args.num_units * args.labels_per_unit
         start, end = sharding.shard_range(total, w, r)
+    if args.num_units is None:
+        ap_err = "either --num-units or --preset is required"
+        raise SystemExit(ap_err)
     opts = gsm_amd.PostSetupOpts(
         data_dir=args.datadir, num_units=args.num_units,
         max_file_size=args.max_file_size, provider_id=args.provider,
@@ -152,7 +164,9 @@ def main():
     i.add_argument("--datadir", required=True)
     i.add_argument("--node-id", required=True)
     i.add_argument("--atx-id", required=True)
-    i.add_argument("--num-units", type=int, required=True)
+    i.add_argument("--num-units", type=int, default=None)
+    i.add_argument("--preset", default=None,
+                   help="mainnet|testnet|fastnet parameter preset")
     i.add_argument("--labels-per-unit", type=int, default=4294967296)
     i.add_argument("--scrypt-n", type=int, default=8192)
     i.add_argument("--max-file-size", type=int, default=4294967296)
