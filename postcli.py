@@ -65,6 +65,8 @@ def cmd_init(args):
     node = bytes.fromhex(args.node_id)
     atx = bytes.fromhex(args.atx_id)
     cfg = _mk_cfg(args)
+    if args.num_units is None:
+        raise SystemExit("either --num-units or --preset is required")
     start = end = 0
     if args.shard:
         import importlib
@@ -72,9 +74,6 @@ def cmd_init(args):
         r, w = (int(x) for x in args.shard.split("/"))
         total = args.num_units * args.labels_per_unit
         start, end = sharding.shard_range(total, w, r)
-    if args.num_units is None:
-        ap_err = "either --num-units or --preset is required"
-        raise SystemExit(ap_err)
     opts = gsm_amd.PostSetupOpts(
         data_dir=args.datadir, num_units=args.num_units,
         max_file_size=args.max_file_size, provider_id=args.provider,
