@@ -1,0 +1,79 @@
+"""Property-based tests (hypothesis) for the pure data-format layers:
+SCALE compact codec, index bit-packing, shard partitioning, PoET merkle
+partial proofs."""
+import ctypes
+import hashlib
+import importlib
+
+from hypothesis import given, settings, strategies as st
+
+wire = importlib.import_module("go-spacemesh_amd.wire")
+sharding = importlib.import_module("go-spacemesh_amd.sharding")
+poet = importlib.import_module("go-spacemesh_amd.poet")
+
+
+@given(st.integers(min_value=0, max_value=2**64 - 1))
+def test_scale_compact_roundtrip(v):
+    enc = wire.encode_compact(v)
+    got, n = wire.decode_compact(enc)
+    assert got == v and n == len(enc)
+
+
+@given(st.integers(min_value=0, max_value=2**32 - 1),
+       st.binary(max_size=800),
+       st.integers(min_value=0, max_value=2**64 - 1))
+def test_postv1_roundtrip(nonce, indices, pw):
+    p = wire.PostV1(nonce=nonce, indices=indices, pow=pw)
+    assert wire.PostV1.decode(p.encode()) == p
+
+
+@given(st.integers(min_value=2, max_value=2**36),
+       st.lists(st.integers(min_value=0), min_size=1, max_size=40))
+@settings(max_examples=50)
+def test_pack_unpack_roundtrip(num_labels, raw):
+    oracle_mod = importlib.import_module("oracle")
+    o = oracle_mod.Oracle()
+    idx = [v % num_labels for v in raw]
+    bpi = o.lib.oracle_bits_per_index(num_labels)
+    k = len(idx)
+    if (k * bpi + 7) // 8 > 800:
+        return
+    arr = (ctypes.c_uint64 * k)(*idx)
+    out = ctypes.create_string_buffer(800)
+    nbytes = o.lib.oracle_pack_indices(arr, k, bpi, out)
+    assert nbytes == (k * bpi + 7) // 8
+    back = (ctypes.c_uint64 * k)()
+    o.lib.oracle_unpack_indices(out.raw, k, bpi, back)
+    assert list(back) == idx
+
+
+@given(st.integers(min_value=0, max_value=2**40),
+       st.integers(min_value=1, max_value=64))
+def test_shard_range_partition(total, world):
+    prev = 0
+    for r in range(world):
+        s, e = sharding.shard_range(total, world, r)
+        assert s == prev and e >= s
+        prev = e
+    assert prev == total
+
+
+@given(st.integers(min_value=0, max_value=6),
+       st.data())
+@settings(max_examples=30, deadline=None)
+def test_poet_partial_proofs(log2n, data):
+    n = 1 << log2n
+    leaves = [hashlib.sha256(b"L%d" % i).digest() for i in range(n)]
+    layers = poet.build_tree(leaves)
+    root = layers[-1][0]
+    k = data.draw(st.integers(min_value=1, max_value=n))
+    idx = sorted(data.draw(st.sets(st.integers(0, n - 1), min_size=k,
+                                   max_size=k)))
+    proof = poet.generate_partial_proof(layers, idx)
+    assert poet.validate_partial_tree(idx, [leaves[i] for i in idx], proof,
+                                      root, n)
+    if proof:
+        bad = list(proof)
+        bad[0] = hashlib.sha256(b"corrupt").digest()
+        assert not poet.validate_partial_tree(
+            idx, [leaves[i] for i in idx], bad, root, n)
