@@ -109,6 +109,11 @@ struct VerifyWorkspace {
   size_t tasks_cap = 0;
   uint32_t *d_cm = nullptr;
   size_t cm_cap = 0; /* words */
+  uint32_t *d_vrk = nullptr; /* 44 words per proof */
+  uint8_t *d_half = nullptr;
+  uint64_t *d_vdiff = nullptr;
+  size_t proofs_cap = 0;
+  uint8_t *d_pass = nullptr; /* per task, with tasks_cap */
 };
 std::mutex g_vws_mu;
 std::map<int, VerifyWorkspace *> g_vws;
@@ -120,6 +125,22 @@ VerifyWorkspace *get_verify_ws(int dev) {
   auto *w = new VerifyWorkspace();
   g_vws[dev] = w;
   return w;
+}
+
+int ws_reserve_proofs(VerifyWorkspace *w, size_t proofs) {
+  if (proofs > w->proofs_cap) {
+    size_t cap = std::max<size_t>(proofs, 1024);
+    if (w->d_vrk) (void)hipFree(w->d_vrk);
+    if (w->d_half) (void)hipFree(w->d_half);
+    if (w->d_vdiff) (void)hipFree(w->d_vdiff);
+    w->d_vrk = nullptr; w->d_half = nullptr; w->d_vdiff = nullptr;
+    w->proofs_cap = 0;
+    HIP_TRY(hipMalloc(&w->d_vrk, cap * 44 * 4));
+    HIP_TRY(hipMalloc(&w->d_half, cap));
+    HIP_TRY(hipMalloc(&w->d_vdiff, cap * 8));
+    w->proofs_cap = cap;
+  }
+  return POST_OK;
 }
 
 int ws_reserve(VerifyWorkspace *w, size_t scratch_bytes, size_t tasks,
@@ -137,13 +158,15 @@ int ws_reserve(VerifyWorkspace *w, size_t scratch_bytes, size_t tasks,
     if (w->d_cid) (void)hipFree(w->d_cid);
     if (w->d_out) (void)hipFree(w->d_out);
     if (w->d_xbuf) (void)hipFree(w->d_xbuf);
+    if (w->d_pass) (void)hipFree(w->d_pass);
     w->d_idx = nullptr; w->d_cid = nullptr; w->d_out = nullptr;
-    w->d_xbuf = nullptr;
+    w->d_xbuf = nullptr; w->d_pass = nullptr;
     w->tasks_cap = 0;
     HIP_TRY(hipMalloc(&w->d_idx, cap * 8));
     HIP_TRY(hipMalloc(&w->d_cid, cap * 4));
     HIP_TRY(hipMalloc(&w->d_out, cap * 32));
     HIP_TRY(hipMalloc(&w->d_xbuf, cap * 128));
+    HIP_TRY(hipMalloc(&w->d_pass, cap));
     w->tasks_cap = cap;
   }
   if (cm_words > w->cm_cap) {
@@ -1230,7 +1253,28 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
   uint64_t lanes = std::min<uint64_t>(
       max_lanes, ((tasks.size() + 127) / 128) * 128);
 
-  std::vector<uint8_t> full((size_t)tasks.size() * 32);
+  /* per-proof cipher data for the device predicate (absolute indexing:
+   * every proof gets a slot, matching task_proof) */
+  std::vector<uint32_t> vrk((size_t)n * 44, 0);
+  std::vector<uint8_t> vhalf(n, 0);
+  std::vector<uint64_t> vdiff(n, 0);
+  for (uint32_t p = 0; p < n; p++) {
+    if (statuses[p] != POST_OK) continue;
+    const PostProof &pr = proofs[p];
+    const PostProofMetadata &me = metas[p];
+    uint8_t key[16];
+    poste::prove_cipher_key(me.challenge, pr.nonce / POSTE_NONCES_PER_AES,
+                            pr.pow, key);
+    poste::aes128_expand(key, vrk.data() + (size_t)p * 44);
+    vhalf[p] = (uint8_t)(pr.nonce % POSTE_NONCES_PER_AES);
+    uint64_t num_labels = (uint64_t)me.num_units * me.labels_per_unit;
+    vdiff[p] = poste::proving_difficulty(cfg->k1, num_labels);
+  }
+  DeviceTables tbl;
+  rc = get_aes_tables((int)cfg->provider_id, tbl);
+  if (rc != POST_OK) return rc;
+
+  std::vector<uint8_t> h_pass(tasks.size());
   const bool dbg = getenv("POST_VERIFY_DEBUG") != nullptr;
   auto tick = [] {
     return std::chrono::duration<double>(
@@ -1275,38 +1319,48 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
     }
     HIP_TRY(poste_launch_label_kernel(&la, (uint32_t)(lanes / 64),
                                       nullptr));
-    HIP_TRY(hipDeviceSynchronize());
     if (dbg) {
-      std::fprintf(stderr, "[verify] kernel %.3fs\n", tick() - t0);
+      (void)hipDeviceSynchronize();
+      std::fprintf(stderr, "[verify] label kernel %.3fs\n", tick() - t0);
       t0 = tick();
     }
-    HIP_TRY(hipMemcpy(full.data(), ws->d_out, full.size(),
+    /* device-side AES threshold predicate over the recomputed labels */
+    rc = ws_reserve_proofs(ws, n);
+    if (rc != POST_OK) return rc;
+    HIP_TRY(hipMemcpy(ws->d_vrk, vrk.data(), vrk.size() * 4,
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(ws->d_half, vhalf.data(), vhalf.size(),
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(ws->d_vdiff, vdiff.data(), vdiff.size() * 8,
+                      hipMemcpyHostToDevice));
+    VerifyPredArgs pa;
+    std::memset(&pa, 0, sizeof(pa));
+    pa.labels2 = (const uint4 *)ws->d_out;
+    pa.count = tasks.size();
+    pa.te = tbl.d_te;
+    pa.sbox = tbl.d_sbox;
+    pa.rk = ws->d_vrk;
+    pa.task_proof = ws->d_cid; /* task->proof map doubles as commit id */
+    pa.half = ws->d_half;
+    pa.difficulty = ws->d_vdiff;
+    pa.pass = ws->d_pass;
+    uint32_t pblocks = (uint32_t)std::min<uint64_t>(
+        (tasks.size() + THREADS - 1) / THREADS, 8192);
+    HIP_TRY(poste_launch_verify_pred_kernel(&pa, pblocks, nullptr));
+    HIP_TRY(hipDeviceSynchronize());
+    if (dbg) {
+      std::fprintf(stderr, "[verify] predicate %.3fs\n", tick() - t0);
+      t0 = tick();
+    }
+    HIP_TRY(hipMemcpy(h_pass.data(), ws->d_pass, h_pass.size(),
                       hipMemcpyDeviceToHost));
   }
 
-  /* final AES threshold predicate on host (cheap; K3*n blocks) */
-  std::vector<std::array<uint32_t, 44>> rk_cache(n);
-  std::vector<bool> rk_ready(n, false);
+  /* collect per-proof verdicts from the device predicate */
   for (size_t i = 0; i < tasks.size(); i++) {
     const Task &t = tasks[i];
     if (statuses[t.proof] != POST_OK) continue;
-    const PostProof &pr = proofs[t.proof];
-    const PostProofMetadata &me = metas[t.proof];
-    if (!rk_ready[t.proof]) {
-      uint8_t key[16];
-      poste::prove_cipher_key(me.challenge,
-                              pr.nonce / POSTE_NONCES_PER_AES, pr.pow, key);
-      poste::aes128_expand(key, rk_cache[t.proof].data());
-      rk_ready[t.proof] = true;
-    }
-    uint8_t out16[16];
-    poste::aes128_enc_block(rk_cache[t.proof].data(), &full[i * 32], out16);
-    uint32_t half = pr.nonce % POSTE_NONCES_PER_AES;
-    uint64_t v = 0;
-    for (int b = 0; b < 8; b++)
-      v |= (uint64_t)out16[8 * half + b] << (8 * b);
-    uint64_t num_labels = (uint64_t)me.num_units * me.labels_per_unit;
-    if (v >= poste::proving_difficulty(cfg->k1, num_labels)) {
+    if (!h_pass[i]) {
       statuses[t.proof] = POST_ERR_INVALID_INDEX;
       if (invalid_indices) invalid_indices[t.proof] = t.position;
     }

@@ -47,9 +47,24 @@ struct ScanKernelArgs {
   uint32_t hit_cap;
 };
 
+struct VerifyPredArgs {
+  const uint4 *labels2;      /* 2 x uint4 per task (full 32-B labels) */
+  uint64_t count;
+  const uint32_t *te;        /* 1024 words */
+  const uint8_t *sbox;       /* 256 bytes */
+  const uint32_t *rk;        /* 44 BE words per PROOF */
+  const uint32_t *task_proof;/* proof index per task */
+  const uint8_t *half;       /* per-proof nonce half (0/1) */
+  const uint64_t *difficulty;/* per-proof u64 threshold */
+  uint8_t *pass;             /* per-task verdict out */
+};
+
 extern "C" {
 hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
                                      uint32_t blocks, hipStream_t stream);
+hipError_t poste_launch_verify_pred_kernel(const VerifyPredArgs *args,
+                                           uint32_t blocks,
+                                           hipStream_t stream);
 hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
                                     uint32_t blocks, hipStream_t stream);
 uint64_t poste_label_resident_slots(uint32_t gap_shift);

@@ -759,6 +759,71 @@ post_scan_kernel(ScanKernelArgs a) {
   }
 }
 
+/* verification's final predicate: AES-encrypt each recomputed label with
+ * its proof's cipher and compare the nonce-half u64 against the proof's
+ * difficulty (the last step of verifying.ProofVerifier.Verify,
+ * post_verifier.go:159 -> per-index AES threshold check). */
+__global__ void __launch_bounds__(POSTE_THREADS)
+post_verify_pred_kernel(VerifyPredArgs a) {
+  extern __shared__ uint32_t lds[];
+  uint32_t *sTe = lds;
+  uint8_t *sSbox = (uint8_t *)(sTe + 1024);
+  for (uint32_t i = threadIdx.x; i < 1024; i += blockDim.x) sTe[i] = a.te[i];
+  for (uint32_t i = threadIdx.x; i < 256; i += blockDim.x)
+    sSbox[i] = a.sbox[i];
+  __syncthreads();
+  const unsigned long long stride =
+      (unsigned long long)gridDim.x * blockDim.x;
+  for (unsigned long long t =
+           (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+       t < a.count; t += stride) {
+    const uint32_t pidx = a.task_proof[t];
+    const uint32_t *rk = a.rk + (unsigned long long)pidx * 44;
+    uint4 lraw = a.labels2[2 * t]; /* first 16 B of the full label */
+    uint32_t w0 = __builtin_bswap32(lraw.x) ^ rk[0];
+    uint32_t w1 = __builtin_bswap32(lraw.y) ^ rk[1];
+    uint32_t w2 = __builtin_bswap32(lraw.z) ^ rk[2];
+    uint32_t w3 = __builtin_bswap32(lraw.w) ^ rk[3];
+#pragma unroll
+    for (int r = 1; r < 10; r++) {
+      uint32_t n0 = sTe[w0 >> 24] ^ sTe[256 + ((w1 >> 16) & 0xff)] ^
+                    sTe[512 + ((w2 >> 8) & 0xff)] ^ sTe[768 + (w3 & 0xff)] ^
+                    rk[4 * r];
+      uint32_t n1 = sTe[w1 >> 24] ^ sTe[256 + ((w2 >> 16) & 0xff)] ^
+                    sTe[512 + ((w3 >> 8) & 0xff)] ^ sTe[768 + (w0 & 0xff)] ^
+                    rk[4 * r + 1];
+      uint32_t n2 = sTe[w2 >> 24] ^ sTe[256 + ((w3 >> 16) & 0xff)] ^
+                    sTe[512 + ((w0 >> 8) & 0xff)] ^ sTe[768 + (w1 & 0xff)] ^
+                    rk[4 * r + 2];
+      uint32_t n3 = sTe[w3 >> 24] ^ sTe[256 + ((w0 >> 16) & 0xff)] ^
+                    sTe[512 + ((w1 >> 8) & 0xff)] ^ sTe[768 + (w2 & 0xff)] ^
+                    rk[4 * r + 3];
+      w0 = n0; w1 = n1; w2 = n2; w3 = n3;
+    }
+    uint32_t f0 = (((uint32_t)sSbox[w0 >> 24] << 24) |
+                   ((uint32_t)sSbox[(w1 >> 16) & 0xff] << 16) |
+                   ((uint32_t)sSbox[(w2 >> 8) & 0xff] << 8) |
+                   sSbox[w3 & 0xff]) ^ rk[40];
+    uint32_t f1 = (((uint32_t)sSbox[w1 >> 24] << 24) |
+                   ((uint32_t)sSbox[(w2 >> 16) & 0xff] << 16) |
+                   ((uint32_t)sSbox[(w3 >> 8) & 0xff] << 8) |
+                   sSbox[w0 & 0xff]) ^ rk[41];
+    uint32_t f2 = (((uint32_t)sSbox[w2 >> 24] << 24) |
+                   ((uint32_t)sSbox[(w3 >> 16) & 0xff] << 16) |
+                   ((uint32_t)sSbox[(w0 >> 8) & 0xff] << 8) |
+                   sSbox[w1 & 0xff]) ^ rk[42];
+    uint32_t f3 = (((uint32_t)sSbox[w3 >> 24] << 24) |
+                   ((uint32_t)sSbox[(w0 >> 16) & 0xff] << 16) |
+                   ((uint32_t)sSbox[(w1 >> 8) & 0xff] << 8) |
+                   sSbox[w2 & 0xff]) ^ rk[43];
+    unsigned long long v =
+        a.half[pidx] == 0
+            ? __builtin_bswap64(((unsigned long long)f0 << 32) | f1)
+            : __builtin_bswap64(((unsigned long long)f2 << 32) | f3);
+    a.pass[t] = v < a.difficulty[pidx] ? 1 : 0;
+  }
+}
+
 /* host-visible launchers (called from engine.cpp) */
 extern "C" {
 
@@ -790,6 +855,15 @@ hipError_t poste_launch_label_kernel(const LabelKernelArgs *args,
 /* Max simultaneously-resident label SLOTS (in-flight labels) of the ROMix
  * kernel for the given gap config.  Scratch beyond this is wasted memory:
  * extra workgroups just queue. */
+hipError_t poste_launch_verify_pred_kernel(const VerifyPredArgs *args,
+                                           uint32_t blocks,
+                                           hipStream_t stream) {
+  size_t lds = 1024 * 4 + 256;
+  hipLaunchKernelGGL(post_verify_pred_kernel, dim3(blocks),
+                     dim3(POSTE_THREADS), lds, stream, *args);
+  return hipGetLastError();
+}
+
 uint64_t poste_label_resident_slots(uint32_t gap_shift) {
   const void *kern =
       gap_shift == 0
