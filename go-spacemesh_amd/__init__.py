@@ -12,7 +12,10 @@ load and run, and missing kernels raise immediately.  The CPU oracle under
 oracle/ is test infrastructure only and is never imported by this package.
 """
 from . import events  # noqa: F401
-from .verifier_pool import OffloadingVerifier  # noqa: F401
+from .verifier_pool import (  # noqa: F401
+    BatchingVerifier,
+    OffloadingVerifier,
+)
 from .api import (  # noqa: F401
     Engine,
     EngineError,
@@ -28,7 +31,7 @@ from .api import (  # noqa: F401
 )
 
 __all__ = [
-    "Engine", "EngineError", "OffloadingVerifier", "PostConfig", "PostProof",
+    "BatchingVerifier", "Engine", "EngineError", "OffloadingVerifier", "PostConfig", "PostProof",
     "PostProofMetadata", "PostSetupManager", "PostSetupOpts", "PostVerifier",
     "ProveOpts", "VerifyOpts", "events", "load_engine",
 ]

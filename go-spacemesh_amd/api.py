@@ -148,6 +148,10 @@ def load_engine() -> ctypes.CDLL:
                                           POINTER(CProofMetadata), c_uint32,
                                           POINTER(_CVerifyConfig),
                                           POINTER(c_int), POINTER(c_uint32)]),
+            "post_verify_batch_seeded": (c_int, [
+                POINTER(CProof), POINTER(CProofMetadata), c_uint32,
+                POINTER(_CVerifyConfig), ctypes.c_char_p, c_size_t,
+                POINTER(c_int), POINTER(c_uint32)]),
             "post_verify_vrf_nonce": (c_int, [POINTER(CProofMetadata),
                                               c_uint64, c_uint32, c_uint32]),
             "post_selftest_blake3": (None, [ctypes.c_char_p, c_size_t,
@@ -494,7 +498,11 @@ class PostVerifier:
                                   f"{inv.value}")
         self.engine._check(rc)
 
-    def verify_batch(self, proofs, metas, opts: VerifyOpts = VerifyOpts()):
+    def verify_batch(self, proofs, metas, opts: VerifyOpts = VerifyOpts(),
+                     seeds=None):
+        # seeds (optional): equal-length per-proof subset seeds — each
+        # gossip verify samples with its own peer seed
+        # (validation.go:206-209).
         n = len(proofs)
         vc, _seed = self._vc(opts)
         cps = (CProof * n)()
@@ -505,8 +513,16 @@ class PostVerifier:
             cms[i] = m.to_c()
         statuses = (c_int * n)()
         invs = (c_uint32 * n)()
-        self.engine._check(self.engine.lib.post_verify_batch(
-            cps, cms, n, byref(vc), statuses, invs))
+        if seeds is not None:
+            assert len(seeds) == n and n > 0
+            slen = len(seeds[0])
+            assert all(len(x) == slen for x in seeds)
+            blob = b"".join(seeds)
+            self.engine._check(self.engine.lib.post_verify_batch_seeded(
+                cps, cms, n, byref(vc), blob, slen, statuses, invs))
+        else:
+            self.engine._check(self.engine.lib.post_verify_batch(
+                cps, cms, n, byref(vc), statuses, invs))
         return [(Status(statuses[i]), invs[i]) for i in range(n)]
 
     def verify_vrf_nonce(self, meta: PostProofMetadata, index: int,

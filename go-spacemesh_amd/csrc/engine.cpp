@@ -1143,6 +1143,15 @@ int post_prove(const char *data_dir, const PostProveConfig *cfg,
 int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
                       uint32_t n, const PostVerifyConfig *cfg, int *statuses,
                       uint32_t *invalid_indices) {
+  return post_verify_batch_seeded(proofs, metas, n, cfg, nullptr, 0,
+                                  statuses, invalid_indices);
+}
+
+int post_verify_batch_seeded(const PostProof *proofs,
+                             const PostProofMetadata *metas, uint32_t n,
+                             const PostVerifyConfig *cfg,
+                             const uint8_t *subset_seeds, size_t seed_len,
+                             int *statuses, uint32_t *invalid_indices) {
   if (!proofs || !metas || !cfg || !statuses) return POST_ERR_INVALID_ARGS;
   if (cfg->pow_mode != POST_POW_MODE_BLAKE3) {
     set_error("RandomX k2pow is not supported (use POST_POW_MODE_BLAKE3)");
@@ -1196,12 +1205,15 @@ int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
                           proof_indices[p].data());
 
     std::vector<uint32_t> positions;
+    const uint8_t *seed = subset_seeds ? subset_seeds + (size_t)p * seed_len
+                                       : cfg->subset_seed;
+    size_t slen = subset_seeds ? seed_len : cfg->subset_seed_len;
     if (cfg->selected_index >= 0) {
       positions.push_back((uint32_t)cfg->selected_index);
-    } else if (cfg->subset_seed && cfg->k3 < cfg->k2) {
+    } else if (seed && cfg->k3 < cfg->k2) {
       positions.resize(cfg->k3);
-      poste::subset_positions(cfg->k2, cfg->k3, cfg->subset_seed,
-                              cfg->subset_seed_len, positions.data());
+      poste::subset_positions(cfg->k2, cfg->k3, seed, slen,
+                              positions.data());
     } else {
       positions.resize(cfg->k2);
       for (uint32_t i = 0; i < cfg->k2; i++) positions[i] = i;

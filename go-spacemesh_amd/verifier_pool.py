@@ -156,3 +156,89 @@ class OffloadingVerifier:
             self._q.put((2, next(self._seq), None))
         for t in list(self._workers):
             t.join(timeout=10)
+
+
+class BatchingVerifier:
+    """MI355X-first variant of the offloading verifier: callers still block
+    per proof (the reference's worker-pool call shape), but a collector
+    groups concurrent jobs into one engine verify_batch call — the engine's
+    natural grain (one label-recompute launch + one predicate launch per
+    batch).  Each job carries its own subset seed, matching
+    the per-peer Subset(k3, seed) of gossip verification
+    (validation.go:206-209).
+
+    inner must expose verify_batch(proofs, metas, opts, seeds) — a
+    PostVerifier does."""
+
+    def __init__(self, inner, max_batch: int = 256,
+                 max_wait_s: float = 0.002, seed_len: int = 32) -> None:
+        self._inner = inner
+        self._max_batch = max_batch
+        self._max_wait = max_wait_s
+        self._seed_len = seed_len
+        self._q: "queue.Queue" = queue.Queue()
+        self._closed = False
+        self._t = threading.Thread(target=self._collector, daemon=True)
+        self._t.start()
+
+    def _collector(self) -> None:
+        import time as _time
+        while True:
+            item = self._q.get()
+            if item is None:
+                return
+            batch = [item]
+            deadline = _time.monotonic() + self._max_wait
+            while len(batch) < self._max_batch:
+                timeout = deadline - _time.monotonic()
+                if timeout <= 0:
+                    break
+                try:
+                    nxt = self._q.get(timeout=timeout)
+                except queue.Empty:
+                    break
+                if nxt is None:
+                    self._run(batch)
+                    return
+                batch.append(nxt)
+            self._run(batch)
+
+    def _run(self, batch) -> None:
+        _metrics.post_verification_queue.set(self._q.qsize())
+        try:
+            res = self._inner.verify_batch(
+                [j[0] for j in batch], [j[1] for j in batch],
+                seeds=[j[2] for j in batch])
+            for (job, (status, inv)) in zip(batch, res):
+                job[3]["status"] = (status, inv)
+                job[4].set()
+        except BaseException as e:  # noqa: BLE001
+            for job in batch:
+                job[3]["error"] = e
+                job[4].set()
+
+    def verify(self, proof, meta, seed: bytes) -> None:
+        """Blocking; raises EngineError(INVALID_INDEX/POW/...) like the
+        single-proof verifier."""
+        if self._closed:
+            raise RuntimeError("verifier closed")
+        if len(seed) != self._seed_len:
+            raise ValueError(f"seed must be {self._seed_len} bytes")
+        import time as _time
+        out: dict = {}
+        done = threading.Event()
+        t0 = _time.monotonic()
+        self._q.put((proof, meta, seed, out, done))
+        done.wait()
+        _metrics.post_verification_latency.observe(_time.monotonic() - t0)
+        if "error" in out:
+            raise out["error"]
+        status, inv = out["status"]
+        if int(status) != 0:
+            from .api import EngineError
+            raise EngineError(int(status), f"position {inv}")
+
+    def close(self) -> None:
+        self._closed = True
+        self._q.put(None)
+        self._t.join(timeout=10)

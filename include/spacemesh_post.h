@@ -239,10 +239,19 @@ int post_verify(const PostProof *proof, const PostProofMetadata *meta,
 /* Batched verification (the verifier-pool steady state, BASELINE config 5):
  * n proofs with per-proof metadata; statuses[i] and invalid_indices[i]
  * receive per-proof results.  One kernel launch recomputes all sampled
- * labels. */
+ * labels and one applies the AES threshold predicate. */
 int post_verify_batch(const PostProof *proofs, const PostProofMetadata *metas,
                       uint32_t n, const PostVerifyConfig *cfg,
                       int *statuses, uint32_t *invalid_indices);
+
+/* Same with PER-PROOF subset seeds (each gossip verification samples with
+ * its own peer-derived seed, validation.go:206-209): subset_seeds holds n
+ * fixed-length seeds back to back (NULL -> cfg->subset_seed for all). */
+int post_verify_batch_seeded(const PostProof *proofs,
+                             const PostProofMetadata *metas, uint32_t n,
+                             const PostVerifyConfig *cfg,
+                             const uint8_t *subset_seeds, size_t seed_len,
+                             int *statuses, uint32_t *invalid_indices);
 
 /* verifying.VerifyVRFNonce (validation.go:261-286). */
 int post_verify_vrf_nonce(const PostProofMetadata *meta, uint64_t index,

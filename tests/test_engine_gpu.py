@@ -645,3 +645,55 @@ def test_verify_batch_commitments_survive_early_rejects(roundtrip):
     assert res[1][0] == gsm_amd.api.Status.OK
     assert res[2][0] == gsm_amd.api.Status.INVALID_ARGS
     assert res[3][0] == gsm_amd.api.Status.OK
+
+
+def test_per_proof_seeds_match_oracle(roundtrip):
+    """Seeded batch verification: each proof sampled with its own seed must
+    agree with the oracle's per-seed verdicts (the gossip shape —
+    per-peer Subset seeds, validation.go:206-209)."""
+    NU, LPU, N, labels, proof, _ = roundtrip
+    o = Oracle()
+    ometa = make_meta(NODE, ATX, CHALLENGE, NU, LPU)
+    cfg = gsm_amd.PostConfig(k1=12, k2=8, k3=3, pow_difficulty=POW_DIFF)
+    ver = gsm_amd.PostVerifier(cfg, scrypt_n=N)
+    meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, NU, LPU)
+    seeds = [bytes([i]) * 32 for i in range(8)]
+    res = ver.verify_batch([proof] * 8, [meta] * 8, seeds=seeds)
+    assert all(s == gsm_amd.api.Status.OK for s, _ in res)
+
+    def o_proof(p):
+        op = Proof()
+        op.nonce = p.nonce
+        op.pow = p.pow
+        op.num_indices = 8
+        op.indices_len = len(p.indices)
+        for i, b in enumerate(p.indices):
+            op.indices[i] = b
+        return op
+
+    # corrupt one byte; per-seed verdicts must match the oracle exactly
+    bad = bytearray(proof.indices)
+    bad[2] ^= 0x3C
+    bp = gsm_amd.PostProof(proof.nonce, bytes(bad), proof.pow)
+    res = ver.verify_batch([bp] * 8, [meta] * 8, seeds=seeds)
+    for seed, (status, inv) in zip(seeds, res):
+        orc, oinv = o.verify(o_proof(bp), ometa, N, 12, 8, 3, seed, -1,
+                             POW_DIFF)
+        assert (orc == 0) == (status == gsm_amd.api.Status.OK), seed
+        if orc == 1:
+            assert inv == oinv
+
+
+def test_batching_verifier_end_to_end_gpu(roundtrip):
+    """BatchingVerifier over the real engine: concurrent per-seed verifies
+    grouped into engine batches."""
+    NU, LPU, N, labels, proof, _ = roundtrip
+    from concurrent.futures import ThreadPoolExecutor
+    cfg = gsm_amd.PostConfig(k1=12, k2=8, k3=3, pow_difficulty=POW_DIFF)
+    ver = gsm_amd.PostVerifier(cfg, scrypt_n=N)
+    meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, NU, LPU)
+    bv = gsm_amd.BatchingVerifier(ver, max_batch=32, max_wait_s=0.01)
+    with ThreadPoolExecutor(8) as ex:
+        list(ex.map(lambda i: bv.verify(proof, meta, bytes([i]) * 32),
+                    range(24)))
+    bv.close()

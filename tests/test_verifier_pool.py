@@ -135,3 +135,45 @@ def test_init_lifecycle_emits_events():
         assert seen == []
     finally:
         unsub()
+
+
+def test_batching_verifier_groups_jobs():
+    calls = []
+
+    class BatchInner:
+        def verify_batch(self, proofs, metas, opts=None, seeds=None):
+            calls.append((len(proofs), seeds))
+            time.sleep(0.01)
+            return [(0, 0) for _ in proofs]
+
+    bv = gsm_amd.BatchingVerifier(BatchInner(), max_batch=64,
+                                  max_wait_s=0.05, seed_len=4)
+    threads = [threading.Thread(target=bv.verify,
+                                args=(i, FakeMeta(), b"s%03d" % i))
+               for i in range(20)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    bv.close()
+    total = sum(n for n, _ in calls)
+    assert total == 20
+    assert len(calls) < 20  # actually batched
+    # per-job seeds forwarded
+    all_seeds = [s for _, seeds in calls for s in seeds]
+    assert sorted(all_seeds) == sorted(b"s%03d" % i for i in range(20))
+
+
+def test_batching_verifier_propagates_failures():
+    class BatchInner:
+        def verify_batch(self, proofs, metas, opts=None, seeds=None):
+            # first proof invalid at position 3, rest OK
+            out = [(0, 0) for _ in proofs]
+            out[0] = (8, 3)  # INVALID_INDEX
+            return out
+
+    bv = gsm_amd.BatchingVerifier(BatchInner(), seed_len=1)
+    with pytest.raises(gsm_amd.EngineError) as e:
+        bv.verify("p", FakeMeta(), b"x")
+    assert e.value.code == gsm_amd.api.Status.INVALID_INDEX
+    bv.close()
