@@ -34,13 +34,20 @@ from gsm_amd import wire  # noqa: E402
 
 
 class Prover:
-    """Engine-backed prover over a postdata dir."""
+    """Engine-backed prover over a postdata dir.  K1/K2/pow-difficulty
+    arrive via argv like the reference's post-service
+    (post_supervisor.go:228-247)."""
 
-    def __init__(self, datadir: str, nonces: int, threads: int) -> None:
+    def __init__(self, datadir: str, nonces: int, threads: int,
+                 k1: int = 26, k2: int = 37,
+                 pow_difficulty: bytes = None) -> None:
         self.datadir = datadir
         self.md = wire.PostMetadata.read(datadir)
         self.nonces = nonces
         self.threads = threads
+        self.k1 = k1
+        self.k2 = k2
+        self.pow_difficulty = pow_difficulty
 
     def metadata(self) -> sp.Metadata:
         return sp.Metadata(node_id=self.md.node_id,
@@ -51,7 +58,9 @@ class Prover:
 
     def prove(self, challenge: bytes) -> sp.Proof:
         cfg = gsm_amd.PostConfig(labels_per_unit=self.md.labels_per_unit,
-                                 min_num_units=1)
+                                 min_num_units=1, k1=self.k1, k2=self.k2)
+        if self.pow_difficulty:
+            cfg.pow_difficulty = self.pow_difficulty
         p = gsm_amd.api.prove_dir(
             self.datadir, challenge, cfg,
             gsm_amd.ProveOpts(nonces=self.nonces, threads=self.threads))
@@ -168,12 +177,18 @@ def main() -> None:
     ap.add_argument("--max-retries", type=int, default=3)
     ap.add_argument("--threads", type=int, default=0)
     ap.add_argument("--nonces", type=int, default=288)
+    ap.add_argument("--k1", type=int, default=26)
+    ap.add_argument("--k2", type=int, default=37)
+    ap.add_argument("--pow-difficulty", default=None)
     ap.add_argument("--mock-prover", action="store_true")
     args = ap.parse_args()
     if args.watch_pid:
         watch_pid(args.watch_pid)
+    pow_diff = bytes.fromhex(args.pow_difficulty) \
+        if args.pow_difficulty else None
     prover = (MockProver(args.dir) if args.mock_prover
-              else Prover(args.dir, args.nonces, args.threads))
+              else Prover(args.dir, args.nonces, args.threads,
+                          args.k1, args.k2, pow_diff))
     print(f"post-service registering at {args.address}", file=sys.stderr)
     serve(args.address, prover, args.max_retries)
 

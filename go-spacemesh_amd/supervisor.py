@@ -159,6 +159,8 @@ class PostSupervisor:
 
     def __init__(self, address: str, datadir: str, nonces: int = 288,
                  threads: int = 0, max_retries: int = 3,
+                 k1: int = 26, k2: int = 37,
+                 pow_difficulty: Optional[bytes] = None,
                  mock_prover: bool = False,
                  on_fatal: Optional[Callable[[int], None]] = None) -> None:
         self.address = address
@@ -168,7 +170,10 @@ class PostSupervisor:
                      "--watch-pid", str(os.getpid()),       # :246
                      "--max-retries", str(max_retries),     # :248-251
                      "--nonces", str(nonces),
+                     "--k1", str(k1), "--k2", str(k2),      # :228-247
                      "--threads", str(threads)]
+        if pow_difficulty:
+            self.argv += ["--pow-difficulty", pow_difficulty.hex()]
         if mock_prover:
             self.argv.append("--mock-prover")
         self._proc: Optional[subprocess.Popen] = None

@@ -46,7 +46,8 @@ def test_full_node_shaped_roundtrip(tmp_path):
 
         # 2. out-of-process prover: supervisor + service child over gRPC
         server = sup_mod.PostServiceServer()
-        sup = sup_mod.PostSupervisor(server.address, d, nonces=16)
+        sup = sup_mod.PostSupervisor(server.address, d, nonces=16,
+                                     k1=12, k2=8, pow_difficulty=POW_DIFF)
         sup.start()
         try:
             client = server.wait_for_client(timeout=30, poll_interval=0.2)

@@ -403,7 +403,7 @@ def test_post_service_real_prover_roundtrip(tmp_path):
     mgr.reset()
 
     server = sup_mod.PostServiceServer()
-    sup = sup_mod.PostSupervisor(server.address, d, nonces=16)
+    sup = sup_mod.PostSupervisor(server.address, d, nonces=16, k1=26, k2=37)
     sup.start()
     try:
         client = server.wait_for_client(timeout=30, poll_interval=0.2)
