@@ -393,6 +393,13 @@ int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
   HIP_TRY(hipStreamCreate(&s->stream));
 
   if (!s->data_dir.empty()) {
+    /* the dir is bound to one identity/config (verifyMetadata role) */
+    rc = check_existing_metadata(s);
+    if (rc != POST_OK) return rc;
+    rc = write_metadata(s); /* persisted at session creation, like
+                               initialization.SaveMetadata; rewritten with
+                               the nonce at completion */
+    if (rc != POST_OK) return rc;
     uint64_t per_file = std::max<uint64_t>(
         1, s->cfg.max_file_size / POST_LABEL_SIZE);
     s->written = existing_labels(s->data_dir, per_file, s->range_start, range);
@@ -464,6 +471,68 @@ static void b64(const uint8_t *in, size_t n, std::string &out) {
     out += i + 1 < n ? t[(v >> 6) & 63] : '=';
     out += i + 2 < n ? t[v & 63] : '=';
   }
+}
+
+/* Minimal reader for the fields write_metadata emits: enough to implement
+ * the reference's metadata guard (initialization.LoadMetadata +
+ * verifyMetadata semantics: a data dir is bound to one identity/config —
+ * activation/post.go:373-435 reads the persisted commitment). */
+static bool md_find_string(const std::string &j, const char *key,
+                           std::string &out) {
+  std::string pat = std::string("\"") + key + "\": \"";
+  size_t p = j.find(pat);
+  if (p == std::string::npos) return false;
+  p += pat.size();
+  size_t e = j.find('"', p);
+  if (e == std::string::npos) return false;
+  out = j.substr(p, e - p);
+  return true;
+}
+
+static bool md_find_u64(const std::string &j, const char *key,
+                        uint64_t &out) {
+  std::string pat = std::string("\"") + key + "\": ";
+  size_t p = j.find(pat);
+  if (p == std::string::npos) return false;
+  out = strtoull(j.c_str() + p + pat.size(), nullptr, 10);
+  return true;
+}
+
+static int check_existing_metadata(PostInitSession *s) {
+  char path[4096];
+  std::snprintf(path, sizeof path, "%s/postdata_metadata.json",
+                s->data_dir.c_str());
+  FILE *f = std::fopen(path, "rb");
+  if (!f) return POST_OK; /* fresh dir */
+  std::string j;
+  char buf[4096];
+  size_t rd;
+  while ((rd = std::fread(buf, 1, sizeof buf, f)) > 0) j.append(buf, rd);
+  std::fclose(f);
+  std::string nid_b64, atx_b64, want_nid, want_atx;
+  b64(s->cfg.node_id, 32, want_nid);
+  b64(s->cfg.commitment_atx_id, 32, want_atx);
+  uint64_t lpu = 0, n = 0;
+  if (md_find_string(j, "NodeId", nid_b64) && nid_b64 != want_nid) {
+    set_error("data dir belongs to a different node id (postdata_metadata"
+              ".json mismatch)");
+    return POST_ERR_INVALID_ARGS;
+  }
+  if (md_find_string(j, "CommitmentAtxId", atx_b64) &&
+      atx_b64 != want_atx) {
+    set_error("data dir was initialized for a different commitment ATX");
+    return POST_ERR_INVALID_ARGS;
+  }
+  if (md_find_u64(j, "LabelsPerUnit", lpu) &&
+      lpu != s->cfg.labels_per_unit) {
+    set_error("data dir was initialized with a different LabelsPerUnit");
+    return POST_ERR_INVALID_ARGS;
+  }
+  if (md_find_u64(j, "N", n) && n != s->cfg.scrypt_n) {
+    set_error("data dir was initialized with different scrypt params");
+    return POST_ERR_INVALID_ARGS;
+  }
+  return POST_OK;
 }
 
 static int write_metadata(PostInitSession *s) {

@@ -697,3 +697,35 @@ def test_batching_verifier_end_to_end_gpu(roundtrip):
         list(ex.map(lambda i: bv.verify(proof, meta, bytes([i]) * 32),
                     range(24)))
     bv.close()
+
+
+def test_datadir_identity_guard(tmp_path):
+    """A postdata dir is bound to one identity/config: re-initializing with
+    a different node id / ATX / params must be refused
+    (initialization metadata guard; commitment read-back at
+    activation/post.go:373-435)."""
+    d = str(tmp_path)
+    cfg, mgr = make_mgr(1, 1 << 10, 32, data_dir=d, max_file_size=1 << 14)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    mgr.reset()
+    # same config resumes fine
+    cfg, mgr2 = make_mgr(1, 1 << 10, 32, data_dir=d, max_file_size=1 << 14)
+    mgr2.prepare_initializer()
+    mgr2.reset()
+    # different identity refused
+    other = gsm_amd.PostSetupManager(
+        bytes([1]) * 32, ATX,
+        gsm_amd.PostConfig(min_num_units=1, labels_per_unit=1 << 10, k1=12,
+                           k2=8, pow_difficulty=POW_DIFF),
+        gsm_amd.PostSetupOpts(num_units=1, scrypt_n=32, data_dir=d,
+                              max_file_size=1 << 14,
+                              scratch_bytes=8 << 30))
+    with pytest.raises(gsm_amd.EngineError) as ei:
+        other.prepare_initializer()
+    assert ei.value.code == gsm_amd.api.Status.INVALID_ARGS
+    # different scrypt N refused
+    cfgn, mgrn = make_mgr(1, 1 << 10, 64, data_dir=d,
+                          max_file_size=1 << 14)
+    with pytest.raises(gsm_amd.EngineError):
+        mgrn.prepare_initializer()
