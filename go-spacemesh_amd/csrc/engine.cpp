@@ -310,6 +310,9 @@ static uint64_t existing_labels(const std::string &dir, uint64_t per_file,
   return done;
 }
 
+static int check_existing_metadata(PostInitSession *s);
+static int write_metadata(PostInitSession *s);
+
 int post_init_new(const PostInitConfig *cfg, PostInitSession **out) {
   if (!cfg || !out) {
     set_error("null args");
