@@ -318,7 +318,6 @@ post_label_prologue_kernel(LabelKernelArgs a) {
     uint32_t hi[8], ho[8], m[16];
     hmac_states_for(cw, index, hi, ho);
     uint32_t Z0[4], Z1[4];
-#pragma unroll
     for (uint32_t half = 0; half < 2; half++) {
       uint32_t cblk[16];
 #pragma unroll
