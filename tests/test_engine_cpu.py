@@ -118,3 +118,16 @@ def test_zero_labels_metadata_rejected(oracle):
     p.num_indices = 8
     rc, _ = oracle.verify(p, meta, 2, 12, 8, 8, None, -1, bytes(32))
     assert rc == 3  # malformed
+
+
+def test_num_units_bounds_enforced():
+    """PostConfig.Min/MaxNumUnits validation (Validator.NumUnits,
+    validation.go:239-249) happens before any GPU work."""
+    cfg = gsm_amd.PostConfig(min_num_units=4, max_num_units=8)
+    for bad in (2, 9):
+        mgr = gsm_amd.PostSetupManager(
+            bytes(32), bytes(32), cfg,
+            gsm_amd.PostSetupOpts(num_units=bad, scrypt_n=2))
+        with pytest.raises(gsm_amd.EngineError) as ei:
+            mgr.prepare_initializer()
+        assert ei.value.code == gsm_amd.api.Status.INVALID_ARGS
