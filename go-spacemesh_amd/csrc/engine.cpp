@@ -631,6 +631,13 @@ int post_init_step(PostInitSession *s, uint64_t max_labels, uint64_t *done) {
     unsigned int n_cand = 0;
     HIP_TRY(hipMemcpyAsync(&n_cand, s->d_cand_count, sizeof(n_cand),
                            hipMemcpyDeviceToHost, s->stream));
+    /* the reference label for this batch's self-check computes on the host
+     * WHILE the kernel runs (a ~5 ms scrypt that would otherwise sit on
+     * the critical path between launches) */
+    const uint64_t probe = count / 2;
+    uint8_t ref[32];
+    const int ref_rc = poste::host_label(s->commitment, args.start + probe,
+                                         s->cfg.scrypt_n, ref);
     HIP_TRY(hipStreamSynchronize(s->stream));
     if (n_cand > 0) {
       unsigned int take = std::min(n_cand, CAND_CAP);
@@ -659,22 +666,16 @@ int post_init_step(PostInitSession *s, uint64_t max_labels, uint64_t *done) {
     }
 
     /* reference-label self-check (ErrReferenceLabelMismatch semantics,
-     * activation/post.go:299-312): recompute one label of the batch on the
-     * host and compare */
-    {
-      uint64_t probe = count / 2;
-      uint8_t ref[32];
-      if (poste::host_label(s->commitment, args.start + probe,
-                            s->cfg.scrypt_n, ref) != 0) {
-        set_error("host reference label failed");
-        return POST_ERR;
-      }
-      if (std::memcmp(ref, s->h_batch + probe * POST_LABEL_SIZE,
-                      POST_LABEL_SIZE) != 0) {
-        set_error("reference label mismatch: device labels diverge from the "
-                  "host reference (ErrReferenceLabelMismatch)");
-        return POST_ERR;
-      }
+     * activation/post.go:299-312): compare the host label computed above */
+    if (ref_rc != 0) {
+      set_error("host reference label failed");
+      return POST_ERR;
+    }
+    if (std::memcmp(ref, s->h_batch + probe * POST_LABEL_SIZE,
+                    POST_LABEL_SIZE) != 0) {
+      set_error("reference label mismatch: device labels diverge from the "
+                "host reference (ErrReferenceLabelMismatch)");
+      return POST_ERR;
     }
 
     if (!s->data_dir.empty()) {
