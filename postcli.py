@@ -95,14 +95,17 @@ def cmd_init(args):
 
     t = threading.Thread(target=progress, daemon=True)
     t.start()
+    resumed_at = st["num_labels_written"]
     t0 = time.time()
     mgr.start_session()
     stop.set()
     dt = time.time() - t0
+    done = total - resumed_at
     nonce = mgr.vrf_nonce()
     print(json.dumps({
-        "labels": total, "seconds": round(dt, 1),
-        "labels_per_sec": round(total / dt, 1) if dt > 0 else None,
+        "labels": total, "labels_this_session": done,
+        "seconds": round(dt, 1),
+        "labels_per_sec": round(done / dt, 1) if dt > 0 and done else None,
         "vrf_nonce": nonce[0] if nonce else None,
     }))
     mgr.reset()
