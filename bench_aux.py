@@ -95,8 +95,10 @@ def bench_verify(n_proofs: int):
                                   pow_difficulty=MAINNET_POW_DIFF)
         ver = gsm_amd.PostVerifier(vcfg, scrypt_n=8192)
         vopts = gsm_amd.VerifyOpts(subset_seed=seed)
-        # warmup
-        ver.verify_batch(proofs[:8], metas[:8], vopts)
+        # warmup at FULL batch size: the cached workspace reserves its
+        # buffers here (a first large hipMalloc costs seconds on some
+        # boxes and must not land in the timed region)
+        ver.verify_batch(proofs, metas, vopts)
         t0 = time.perf_counter()
         res = ver.verify_batch(proofs, metas, vopts)
         dt = time.perf_counter() - t0
