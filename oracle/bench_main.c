@@ -71,6 +71,7 @@ static int cmd_bench(int argc, char **argv) {
   if (threads > 0) omp_set_num_threads(threads);
   int used = threads > 0 ? threads : omp_get_max_threads();
 #else
+  (void)threads;
   int used = 1;
 #endif
   uint8_t node_id[32], atx[32], commitment[32];
