@@ -645,7 +645,14 @@ post_scan_kernel(ScanKernelArgs a) {
   /* T-tables + sbox in LDS (random per-lane indices); round keys stay in
    * global — the cipher loop is wave-uniform, so they compile to scalar
    * loads through the constant cache instead of ~5.8K extra LDS reads per
-   * label. */
+   * label.  The kernel is LDS-conflict-throughput bound (ILP 1/2/4 all
+   * measure ~274 M labels/s, profiles/r01_scan_sweep.md); the
+   * POSTE_SCAN_GLOBAL_TT A/B variant reads the tables through the vector
+   * L1 instead, to compare cache-port against LDS-bank serialization. */
+#ifdef POSTE_SCAN_GLOBAL_TT
+  const uint32_t *sTe = a.te;
+  const uint8_t *sSbox = a.sbox;
+#else
   extern __shared__ uint32_t lds[];
   uint32_t *sTe = lds;                      /* 1024 words */
   uint8_t *sSbox = (uint8_t *)(sTe + 1024); /* 256 bytes */
@@ -654,6 +661,7 @@ post_scan_kernel(ScanKernelArgs a) {
   for (uint32_t i = threadIdx.x; i < 256; i += blockDim.x)
     sSbox[i] = a.sbox[i];
   __syncthreads();
+#endif
 
   /* Four independent labels per lane (ILP): a single AES chain leaves the
    * wave ~76% latency-stalled (profiles: SQ_ACTIVE_INST 10%, LDS 14%);
