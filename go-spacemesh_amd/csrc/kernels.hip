@@ -646,16 +646,32 @@ post_scan_kernel(ScanKernelArgs a) {
    * global — the cipher loop is wave-uniform, so they compile to scalar
    * loads through the constant cache instead of ~5.8K extra LDS reads per
    * label.  The kernel is LDS-conflict-throughput bound (ILP 1/2/4 all
-   * measure ~274 M labels/s, profiles/r01_scan_sweep.md); the
-   * POSTE_SCAN_GLOBAL_TT A/B variant reads the tables through the vector
-   * L1 instead, to compare cache-port against LDS-bank serialization. */
-#ifdef POSTE_SCAN_GLOBAL_TT
-  const uint32_t *sTe = a.te;
+   * measure ~274 M labels/s; a random 32-lane b32 gather costs
+   * 2 x E[max 32-into-32-bank load] ~ 6.4 LDS cycles — see
+   * profiles/r01_scan_sweep.md and MI355X_MICROARCH.md §LDS).  Two
+   * gated A/B variants attack that wall (both UNVERIFIED on hardware,
+   * round-2 retest):
+   *   POSTE_SCAN_GLOBAL_TT — all lookups through the vector L1;
+   *   POSTE_SCAN_SPLIT_TT — Te0/Te1 from LDS, Te2/Te3 through the L1,
+   *     so the two memory pipes each carry half the gathers. */
+#if defined(POSTE_SCAN_GLOBAL_TT)
+  const uint32_t *TE_LO = a.te; /* Te0/Te1 (offsets 0, 256) */
+  const uint32_t *TE_HI = a.te; /* Te2/Te3 (offsets 512, 768) */
   const uint8_t *sSbox = a.sbox;
+#elif defined(POSTE_SCAN_SPLIT_TT)
+  extern __shared__ uint32_t lds[];
+  uint32_t *TE_LO = lds;                     /* Te0/Te1: 512 words */
+  const uint32_t *TE_HI = a.te;              /* Te2/Te3 via L1 */
+  uint8_t *sSbox = (uint8_t *)(lds + 512);   /* 256 bytes */
+  for (uint32_t i = threadIdx.x; i < 512; i += blockDim.x) TE_LO[i] = a.te[i];
+  for (uint32_t i = threadIdx.x; i < 256; i += blockDim.x)
+    sSbox[i] = a.sbox[i];
+  __syncthreads();
 #else
   extern __shared__ uint32_t lds[];
   uint32_t *sTe = lds;                      /* 1024 words */
   uint8_t *sSbox = (uint8_t *)(sTe + 1024); /* 256 bytes */
+  uint32_t *TE_LO = sTe, *TE_HI = sTe;
 
   for (uint32_t i = threadIdx.x; i < 1024; i += blockDim.x) sTe[i] = a.te[i];
   for (uint32_t i = threadIdx.x; i < 256; i += blockDim.x)
@@ -705,22 +721,22 @@ post_scan_kernel(ScanKernelArgs a) {
       for (int r = 1; r < 10; r++) {
 #pragma unroll
         for (int u = 0; u < L; u++) {
-          uint32_t n0 = sTe[w[u][0] >> 24] ^
-                        sTe[256 + ((w[u][1] >> 16) & 0xff)] ^
-                        sTe[512 + ((w[u][2] >> 8) & 0xff)] ^
-                        sTe[768 + (w[u][3] & 0xff)] ^ rk[4 * r];
-          uint32_t n1 = sTe[w[u][1] >> 24] ^
-                        sTe[256 + ((w[u][2] >> 16) & 0xff)] ^
-                        sTe[512 + ((w[u][3] >> 8) & 0xff)] ^
-                        sTe[768 + (w[u][0] & 0xff)] ^ rk[4 * r + 1];
-          uint32_t n2 = sTe[w[u][2] >> 24] ^
-                        sTe[256 + ((w[u][3] >> 16) & 0xff)] ^
-                        sTe[512 + ((w[u][0] >> 8) & 0xff)] ^
-                        sTe[768 + (w[u][1] & 0xff)] ^ rk[4 * r + 2];
-          uint32_t n3 = sTe[w[u][3] >> 24] ^
-                        sTe[256 + ((w[u][0] >> 16) & 0xff)] ^
-                        sTe[512 + ((w[u][1] >> 8) & 0xff)] ^
-                        sTe[768 + (w[u][2] & 0xff)] ^ rk[4 * r + 3];
+          uint32_t n0 = TE_LO[w[u][0] >> 24] ^
+                        TE_LO[256 + ((w[u][1] >> 16) & 0xff)] ^
+                        TE_HI[512 + ((w[u][2] >> 8) & 0xff)] ^
+                        TE_HI[768 + (w[u][3] & 0xff)] ^ rk[4 * r];
+          uint32_t n1 = TE_LO[w[u][1] >> 24] ^
+                        TE_LO[256 + ((w[u][2] >> 16) & 0xff)] ^
+                        TE_HI[512 + ((w[u][3] >> 8) & 0xff)] ^
+                        TE_HI[768 + (w[u][0] & 0xff)] ^ rk[4 * r + 1];
+          uint32_t n2 = TE_LO[w[u][2] >> 24] ^
+                        TE_LO[256 + ((w[u][3] >> 16) & 0xff)] ^
+                        TE_HI[512 + ((w[u][0] >> 8) & 0xff)] ^
+                        TE_HI[768 + (w[u][1] & 0xff)] ^ rk[4 * r + 2];
+          uint32_t n3 = TE_LO[w[u][3] >> 24] ^
+                        TE_LO[256 + ((w[u][0] >> 16) & 0xff)] ^
+                        TE_HI[512 + ((w[u][1] >> 8) & 0xff)] ^
+                        TE_HI[768 + (w[u][2] & 0xff)] ^ rk[4 * r + 3];
           w[u][0] = n0; w[u][1] = n1; w[u][2] = n2; w[u][3] = n3;
         }
       }
