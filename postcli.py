@@ -34,8 +34,13 @@ from gsm_amd import wire  # noqa: E402
 
 
 def cmd_providers(_args):
-    for p in gsm_amd.Engine().providers():
+    provs = list(gsm_amd.Engine().providers())
+    for p in provs:
         print(json.dumps(p))
+    if not provs:
+        print("no MI355X providers found (no GPU visible)", file=sys.stderr)
+        return 1
+    return 0
 
 
 def cmd_benchmark(args):
@@ -190,7 +195,7 @@ def main():
     v.add_argument("--pow-difficulty", default=None)
     v.set_defaults(fn=cmd_verify)
     args = ap.parse_args()
-    args.fn(args)
+    sys.exit(args.fn(args))
 
 
 if __name__ == "__main__":
