@@ -77,3 +77,57 @@ def test_poet_partial_proofs(log2n, data):
         bad[0] = hashlib.sha256(b"corrupt").digest()
         assert not poet.validate_partial_tree(
             idx, [leaves[i] for i in idx], bad, root, n)
+
+
+# --- protocol difficulty formulas vs Python big-int arithmetic ----------
+
+@given(st.integers(min_value=1, max_value=2**32 - 1),
+       st.integers(min_value=1, max_value=2**62))
+def test_proving_difficulty_matches_bigint(k1, num_labels):
+    oracle_mod = importlib.import_module("oracle")
+    o = oracle_mod.Oracle()
+    want = min((k1 << 64) // num_labels, 2**64 - 1)
+    assert o.lib.oracle_proving_difficulty(k1, num_labels) == want
+
+
+@given(st.integers(min_value=1, max_value=2**62))
+def test_vrf_difficulty_matches_bigint(num_labels):
+    oracle_mod = importlib.import_module("oracle")
+    o = oracle_mod.Oracle()
+    out = ctypes.create_string_buffer(32)
+    o.lib.oracle_vrf_difficulty(num_labels, out)
+    want = min((16 << 256) // num_labels, 2**256 - 1)
+    assert int.from_bytes(out.raw, "big") == want
+
+
+# --- randomized CPU prove->verify round trip (shrunk parameters) --------
+
+@given(st.binary(min_size=32, max_size=32),
+       st.binary(min_size=32, max_size=32),
+       st.binary(min_size=32, max_size=32),
+       st.integers(min_value=1, max_value=8),
+       st.binary(min_size=1, max_size=16),
+       st.data())
+@settings(max_examples=5, deadline=None)
+def test_oracle_roundtrip_randomized(node, atx, challenge, k3, seed, data):
+    oracle_mod = importlib.import_module("oracle")
+    o = oracle_mod.Oracle()
+    NU, LPU, N, K1, K2 = 1, 128, 2, 12, 8
+    pow_diff = bytes([0x0F]) + bytes([0xFF]) * 31
+    commitment = o.commitment(node, atx)
+    labels, _ = o.init_range(commitment, 0, NU * LPU, N)
+    try:
+        proof = o.prove(labels, NU * LPU, challenge, K1, K2, 64, pow_diff)
+    except ValueError:
+        return  # no qualifying nonce for this instance: legal outcome
+    meta = oracle_mod.make_meta(node, atx, challenge, NU, LPU)
+    rc, _ = o.verify(proof, meta, N, K1, K2, K2, None, -1, pow_diff)
+    assert rc == 0
+    rc, _ = o.verify(proof, meta, N, K1, K2, min(k3, K2), seed, -1, pow_diff)
+    assert rc == 0
+    # corrupt one random byte of the packed indices -> must not verify OK
+    bad = oracle_mod.Proof.from_buffer_copy(proof)
+    pos = data.draw(st.integers(0, max(0, bad.indices_len - 1)))
+    bad.indices[pos] ^= data.draw(st.integers(1, 255))
+    rc, _ = o.verify(bad, meta, N, K1, K2, K2, None, -1, pow_diff)
+    assert rc != 0
