@@ -206,6 +206,40 @@ __global__ void __launch_bounds__(THREADS) k_gather4(uint32_t n,
   if (acc == 0xdeadbeef) sink[0] = acc;
 }
 
+__global__ void __launch_bounds__(THREADS) k_gather8(uint32_t n,
+                                                     uint32_t iters,
+                                                     const uint4 *V,
+                                                     uint32_t *sink) {
+  /* 8 lanes share one 128-B block, one 16-B uint4 per lane at consecutive
+   * addresses -> each wave instruction is 8 aligned 128-B transactions.
+   * Round-2 hypothesis test: do 128-B requests beat the 64-B pattern
+   * (7.7 TB/s measured) enough to justify an oct-cooperative labeling
+   * kernel? */
+  const unsigned long long lane =
+      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned long long lanes =
+      (unsigned long long)gridDim.x * blockDim.x;
+  const unsigned long long group = lane >> 3;
+  const uint32_t sub = (uint32_t)(lane & 7);
+  const unsigned long long groups = lanes >> 3;
+  const uint32_t mask = n - 1;
+  uint32_t acc = (uint32_t)group * 2654435761u;
+  for (uint32_t i = 0; i < iters; i++) {
+    uint32_t j = acc & mask;
+    /* block of group g stored as 8 consecutive uint4 at (j*groups + g) */
+    const uint4 *p = V + ((unsigned long long)j * groups + group) * 8ull;
+    uint4 a = p[sub];
+    uint32_t x = a.x ^ a.y ^ a.z ^ a.w;
+    /* xor-reduce across the 8 lanes so acc stays group-uniform */
+    x ^= __shfl_xor(x, 1);
+    x ^= __shfl_xor(x, 2);
+    x ^= __shfl_xor(x, 4);
+    acc ^= x;
+    acc = acc * 1664525u + 1013904223u;
+  }
+  if (acc == 0xdeadbeef) sink[0] = acc;
+}
+
 int main(int argc, char **argv) {
   uint32_t n = argc > 1 ? (uint32_t)atoi(argv[1]) : 8192;
   uint32_t blocks = argc > 2 ? (uint32_t)atoi(argv[2]) : 768;
@@ -240,7 +274,8 @@ int main(int argc, char **argv) {
                   {"read(phase2 random)", 2, 128.0},
                   {"full(romix)", 3, 256.0},
                   {"gather1(16B reqs)", 4, 128.0},
-                  {"gather4(64B reqs)", 5, 32.0}};
+                  {"gather4(64B reqs)", 5, 32.0},
+                  {"gather8(128B reqs)", 6, 16.0}};
   for (auto &c : cases) {
     /* warmup + timed */
     for (int rep = 0; rep < 2; rep++) {
@@ -270,6 +305,10 @@ int main(int argc, char **argv) {
         hipLaunchKernelGGL(k_gather4, dim3(blocks), dim3(THREADS), 0, 0, n,
                            4 * n, V, sink);
         break;
+      case 6:
+        hipLaunchKernelGGL(k_gather8, dim3(blocks), dim3(THREADS), 0, 0, n,
+                           8 * n, V, sink);
+        break;
       }
       (void)hipEventRecord(e1, nullptr);
       if (hipEventSynchronize(e1) != hipSuccess) {
@@ -280,9 +319,10 @@ int main(int argc, char **argv) {
       float ms = 0;
       (void)hipEventElapsedTime(&ms, e0, e1);
       double iters = (double)lanes * n *
-                     (c.which == 0 || c.which == 3 ? 2.0
-                                                   : (c.which == 5 ? 4.0
-                                                                   : 1.0));
+                     (c.which == 0 || c.which == 3
+                          ? 2.0
+                          : (c.which == 5 ? 4.0
+                                          : (c.which == 6 ? 8.0 : 1.0)));
       double gops = iters * 2.0 * 416 / (ms / 1e3) / 1e9;
       double gbs = iters * c.bytes_per_iter / (ms / 1e3) / 1e9;
       double cyc_per_iter = (ms / 1e3) * 2.4e9 / ((double)n *
