@@ -125,9 +125,18 @@ def test_oracle_roundtrip_randomized(node, atx, challenge, k3, seed, data):
     assert rc == 0
     rc, _ = o.verify(proof, meta, N, K1, K2, min(k3, K2), seed, -1, pow_diff)
     assert rc == 0
-    # corrupt one random byte of the packed indices -> must not verify OK
+    # corrupt one random byte of the packed indices: the verdict is
+    # PROBABILISTIC by design (a flipped index still passes the AES
+    # threshold w.p. ~k1/num_labels — the reference's verifier has the
+    # same property), so assert determinism, not rejection
     bad = oracle_mod.Proof.from_buffer_copy(proof)
     pos = data.draw(st.integers(0, max(0, bad.indices_len - 1)))
     bad.indices[pos] ^= data.draw(st.integers(1, 255))
-    rc, _ = o.verify(bad, meta, N, K1, K2, K2, None, -1, pow_diff)
+    rc1, _ = o.verify(bad, meta, N, K1, K2, K2, None, -1, pow_diff)
+    rc2, _ = o.verify(bad, meta, N, K1, K2, K2, None, -1, pow_diff)
+    assert rc1 == rc2 and rc1 in (0, 1, 3)
+    # guaranteed-negative case: truncated index bytes are malformed
+    short = oracle_mod.Proof.from_buffer_copy(proof)
+    short.indices_len -= 1
+    rc, _ = o.verify(short, meta, N, K1, K2, K2, None, -1, pow_diff)
     assert rc != 0
