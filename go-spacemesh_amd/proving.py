@@ -9,7 +9,7 @@ cipher keys derive from it)."""
 from __future__ import annotations
 
 import ctypes
-from ctypes import POINTER, byref, c_uint32, c_uint64, c_uint8
+from ctypes import POINTER, byref, c_uint32, c_uint64
 from typing import Dict, List, Optional, Sequence, Tuple
 
 from . import api as _api

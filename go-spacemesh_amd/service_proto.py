@@ -16,7 +16,7 @@ v1.55.0 is importable).  Message CONTENT follows the call sites cited above.
 from __future__ import annotations
 
 import dataclasses
-from typing import List, Optional, Tuple
+from typing import Optional, Tuple
 
 
 # ---------------- protobuf wire primitives ----------------

@@ -10,7 +10,7 @@ from __future__ import annotations
 import ctypes
 import os
 import subprocess
-from ctypes import (POINTER, byref, c_int, c_int32, c_int64, c_size_t,
+from ctypes import (POINTER, byref, c_int, c_int32, c_size_t,
                     c_uint8, c_uint16, c_uint32, c_uint64, create_string_buffer)
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
