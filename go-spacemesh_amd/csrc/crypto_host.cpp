@@ -1,6 +1,7 @@
 /* crypto_host.cpp — engine host crypto (see crypto_host.h).  Independent
  * implementation #2; cross-checked in tests against the oracle and the
  * committed OpenSSL/FIPS/BLAKE3 golden vectors. */
+#include <cassert>
 #include "crypto_host.h"
 #include "post_common.h"
 
@@ -70,7 +71,11 @@ static void compress(const uint32_t h[8], const uint32_t m_in[16], uint64_t t,
 } // namespace b3
 
 void blake3(const uint8_t *msg, size_t len, uint8_t *out, size_t outlen) {
-  /* chain full 64-byte blocks, keep the last for the root compressions */
+  /* chain full 64-byte blocks, keep the last for the root compressions;
+   * single-chunk only — beyond 1024 B the BLAKE3 chunk tree would be
+   * required and this would silently diverge, so refuse loudly (every
+   * input on this path is <= 64 B) */
+  assert(len <= 1024 && "single-chunk blake3");
   uint32_t h[8];
   std::memcpy(h, b3::IV, 32);
   size_t nblocks = len == 0 ? 1 : (len + 63) / 64;
