@@ -476,6 +476,34 @@ static void b64(const uint8_t *in, size_t n, std::string &out) {
   }
 }
 
+static bool b64_decode(const std::string &in, uint8_t *out, size_t want) {
+  static int8_t rev[256];
+  static bool init = false;
+  if (!init) {
+    static const char t[] =
+        "ABCDEFGHIJKLMNOPQRSTUVWXYZabcdefghijklmnopqrstuvwxyz0123456789+/";
+    for (int i = 0; i < 256; i++) rev[i] = -1;
+    for (int i = 0; i < 64; i++) rev[(uint8_t)t[i]] = (int8_t)i;
+    init = true;
+  }
+  size_t n = 0;
+  uint32_t acc = 0;
+  int bits = 0;
+  for (char c : in) {
+    if (c == '=') break;
+    int8_t v = rev[(uint8_t)c];
+    if (v < 0) return false;
+    acc = (acc << 6) | (uint32_t)v;
+    bits += 6;
+    if (bits >= 8) {
+      bits -= 8;
+      if (n >= want) return false;
+      out[n++] = (uint8_t)(acc >> bits);
+    }
+  }
+  return n == want;
+}
+
 /* Minimal reader for the fields write_metadata emits: enough to implement
  * the reference's metadata guard (initialization.LoadMetadata +
  * verifyMetadata semantics: a data dir is bound to one identity/config —
@@ -534,6 +562,26 @@ static int check_existing_metadata(PostInitSession *s) {
   if (md_find_u64(j, "N", n) && n != s->cfg.scrypt_n) {
     set_error("data dir was initialized with different scrypt params");
     return POST_ERR_INVALID_ARGS;
+  }
+  /* Resume must carry the persisted VRF nonce forward: the reference
+   * initializer keeps metadata's Nonce/NonceValue across sessions, and a
+   * sharded (index_start/end) session that loses its shard-local minimum
+   * would corrupt the cross-GPU min-reduce after kill+resume.  Seed the
+   * session's running minimum from the file before write_metadata (called
+   * right after us) rewrites it. */
+  uint64_t nonce_idx = 0;
+  std::string nv_b64;
+  if (md_find_u64(j, "Nonce", nonce_idx) &&
+      md_find_string(j, "NonceValue", nv_b64)) {
+    uint8_t label[32];
+    if (!b64_decode(nv_b64, label, 32)) {
+      set_error("postdata_metadata.json NonceValue is not 32 base64 bytes");
+      return POST_ERR_INVALID_ARGS;
+    }
+    std::lock_guard<std::mutex> lk(s->nonce_mu);
+    s->nonce_found = true;
+    s->nonce_idx = nonce_idx;
+    std::memcpy(s->nonce_label, label, 32);
   }
   return POST_OK;
 }

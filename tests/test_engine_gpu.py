@@ -729,3 +729,42 @@ def test_datadir_identity_guard(tmp_path):
                           max_file_size=1 << 14)
     with pytest.raises(gsm_amd.EngineError):
         mgrn.prepare_initializer()
+
+
+def test_nonce_survives_kill_and_resume(tmp_path):
+    """The persisted VRF nonce must survive reopening the data dir: a
+    sharded session's shard-local minimum feeds the cross-GPU min-reduce
+    and cannot be recomputed from the remaining range alone (reference
+    initializer keeps metadata Nonce across resume)."""
+    d = str(tmp_path)
+    lpu = 1 << 12
+    # shard [0, lpu/2): complete it, so the shard's minimum is final
+    cfg, mgr = make_mgr(1, lpu, 128, data_dir=d, max_file_size=1 << 15,
+                        index_start=0, index_end=lpu // 2)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    got = mgr.vrf_nonce()
+    assert got is not None
+    mgr.reset()
+    md = json.load(open(os.path.join(d, "postdata_metadata.json")))
+    assert md["Nonce"] == got[0]
+
+    # reopen the same shard: prepare_initializer alone (before any new
+    # labeling) must already carry the persisted nonce, and metadata must
+    # not lose it to the session-creation rewrite
+    cfg, mgr2 = make_mgr(1, lpu, 128, data_dir=d, max_file_size=1 << 15,
+                         index_start=0, index_end=lpu // 2)
+    mgr2.prepare_initializer()
+    md2 = json.load(open(os.path.join(d, "postdata_metadata.json")))
+    assert md2.get("Nonce") == got[0]
+    assert md2.get("NonceValue") == md["NonceValue"]
+    got2 = mgr2.vrf_nonce()
+    assert got2 == got
+    mgr2.reset()
+
+    # oracle cross-check: the persisted minimum is the true shard minimum
+    o = Oracle()
+    commit = o.commitment(NODE, ATX)
+    _, best = o.init_range(commit, 0, lpu // 2, 128)
+    assert got[0] == best.index
+    assert got[1] == bytes(best.label)
