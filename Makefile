@@ -30,6 +30,12 @@ go-spacemesh_amd/libpost_hip_scan4.so: $(SRCS_ENGINE) $(HDRS_ENGINE)
 scan4: go-spacemesh_amd/libpost_hip_scan4.so
 .PHONY: scan4
 
+# A/B variant: scan ILP depth 1
+go-spacemesh_amd/libpost_hip_scan1.so: $(SRCS_ENGINE) $(HDRS_ENGINE)
+	$(HIPCC) $(HIPFLAGS) -DPOSTE_SCAN_ILP=1 -shared $(SRCS_ENGINE) -o $@
+scan1: go-spacemesh_amd/libpost_hip_scan1.so
+.PHONY: scan1
+
 oracle:
 	$(MAKE) -C oracle
 

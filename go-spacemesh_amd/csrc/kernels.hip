@@ -21,6 +21,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdlib>
+#include <cstring>
 
 #include "kernel_args.h"
 #include "post_common.h"
@@ -782,6 +783,137 @@ post_scan_kernel(ScanKernelArgs a) {
   }
 }
 
+/* Bank-replicated scan: one Te0 table replicated across the 32 LDS banks.
+ * Copy c is strided so element x of copy c sits at word x*32 + c; the bank
+ * of word w is w%32 = c, and each lane reads only copy (lane%32), so every
+ * ds_read_b32 gather in the cipher loop is conflict-free BY CONSTRUCTION
+ * (2 LDS cycles per wave-gather, MI355X_MICROARCH §LDS, vs ~6.4 measured
+ * for the shared-table random gather — profiles/r01_scan_sweep.md).  The
+ * round-2 TT retest (profiles/r02_scan_tt_verdict.md) proved the L1 pipe
+ * cannot bypass that wall, so this attacks it inside the LDS instead:
+ * 3.2x fewer LDS cycles for 32 KB of LDS per workgroup and a few extra
+ * VALU ops — Te1/2/3 derive from Te0 by byte rotation (one v_alignbit
+ * each; te[256+x] = ror8(te[x]), crypto_host.cpp:332-334), and the final
+ * round's S-box is byte 1 of Te0 (te[x] packs [s2,s,s,s3] so s =
+ * (te[x]>>8)&0xff) — no separate sbox table at all. */
+#define ROR8(x) __builtin_amdgcn_alignbit((x), (x), 8)
+#define ROR16(x) __builtin_amdgcn_alignbit((x), (x), 16)
+#define ROR24(x) __builtin_amdgcn_alignbit((x), (x), 24)
+
+__global__ void __launch_bounds__(POSTE_THREADS)
+post_scan_bankrep_kernel(ScanKernelArgs a) {
+  extern __shared__ uint32_t sTe[]; /* 8192 words = 32 copies x 1 KiB */
+  for (uint32_t i = threadIdx.x; i < 8192; i += blockDim.x)
+    sTe[i] = a.te[i >> 5]; /* consecutive lanes -> consecutive banks */
+  __syncthreads();
+  const uint32_t lane31 = threadIdx.x & 31;
+#define TE0R(idx) sTe[(((uint32_t)(idx)) << 5) | lane31]
+
+  const unsigned long long stride =
+      (unsigned long long)gridDim.x * blockDim.x;
+#ifndef POSTE_SCAN_ILP
+#define POSTE_SCAN_ILP 2
+#endif
+  constexpr int L = POSTE_SCAN_ILP;
+  const unsigned long long span = stride * L;
+  for (unsigned long long t0 =
+           (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+       t0 < a.count; t0 += span) {
+    uint32_t p[L][4];
+    unsigned long long ts[L];
+#pragma unroll
+    for (int u = 0; u < L; u++) {
+      unsigned long long t = t0 + (unsigned long long)u * stride;
+      ts[u] = t < a.count ? t : t0; /* clamp: duplicate work, hits only
+                                       recorded for live chains below */
+      uint4 lraw = a.labels[ts[u]];
+      p[u][0] = __builtin_bswap32(lraw.x);
+      p[u][1] = __builtin_bswap32(lraw.y);
+      p[u][2] = __builtin_bswap32(lraw.z);
+      p[u][3] = __builtin_bswap32(lraw.w);
+    }
+    const int live = (int)((a.count - t0 + stride - 1) / stride) < L
+                         ? (int)((a.count - t0 + stride - 1) / stride)
+                         : L;
+    for (uint32_t c = 0; c < a.n_ciphers; c++) {
+      const uint32_t *rk = a.rk + c * 44;
+      uint32_t w[L][4];
+#pragma unroll
+      for (int u = 0; u < L; u++)
+#pragma unroll
+        for (int k = 0; k < 4; k++) w[u][k] = p[u][k] ^ rk[k];
+#pragma unroll
+      for (int r = 1; r < 10; r++) {
+#pragma unroll
+        for (int u = 0; u < L; u++) {
+          uint32_t n0 = TE0R(w[u][0] >> 24) ^
+                        ROR8(TE0R((w[u][1] >> 16) & 0xff)) ^
+                        ROR16(TE0R((w[u][2] >> 8) & 0xff)) ^
+                        ROR24(TE0R(w[u][3] & 0xff)) ^ rk[4 * r];
+          uint32_t n1 = TE0R(w[u][1] >> 24) ^
+                        ROR8(TE0R((w[u][2] >> 16) & 0xff)) ^
+                        ROR16(TE0R((w[u][3] >> 8) & 0xff)) ^
+                        ROR24(TE0R(w[u][0] & 0xff)) ^ rk[4 * r + 1];
+          uint32_t n2 = TE0R(w[u][2] >> 24) ^
+                        ROR8(TE0R((w[u][3] >> 16) & 0xff)) ^
+                        ROR16(TE0R((w[u][0] >> 8) & 0xff)) ^
+                        ROR24(TE0R(w[u][1] & 0xff)) ^ rk[4 * r + 2];
+          uint32_t n3 = TE0R(w[u][3] >> 24) ^
+                        ROR8(TE0R((w[u][0] >> 16) & 0xff)) ^
+                        ROR16(TE0R((w[u][1] >> 8) & 0xff)) ^
+                        ROR24(TE0R(w[u][2] & 0xff)) ^ rk[4 * r + 3];
+          w[u][0] = n0; w[u][1] = n1; w[u][2] = n2; w[u][3] = n3;
+        }
+      }
+      /* final round: S[x] = (Te0[x]>>8)&0xff; assemble each output word
+       * from four replicated-table gathers (still conflict-free) */
+#define SB24(idx) ((TE0R(idx) & 0xff00u) << 16)
+#define SB16(idx) ((TE0R(idx) & 0xff00u) << 8)
+#define SB08(idx) (TE0R(idx) & 0xff00u)
+#define SB00(idx) ((TE0R(idx) >> 8) & 0xffu)
+#pragma unroll
+      for (int u = 0; u < L; u++) {
+        if (u >= live) break;
+        uint32_t f0 = (SB24(w[u][0] >> 24) | SB16((w[u][1] >> 16) & 0xff) |
+                       SB08((w[u][2] >> 8) & 0xff) | SB00(w[u][3] & 0xff)) ^
+                      rk[40];
+        uint32_t f1 = (SB24(w[u][1] >> 24) | SB16((w[u][2] >> 16) & 0xff) |
+                       SB08((w[u][3] >> 8) & 0xff) | SB00(w[u][0] & 0xff)) ^
+                      rk[41];
+        uint32_t f2 = (SB24(w[u][2] >> 24) | SB16((w[u][3] >> 16) & 0xff) |
+                       SB08((w[u][0] >> 8) & 0xff) | SB00(w[u][1] & 0xff)) ^
+                      rk[42];
+        uint32_t f3 = (SB24(w[u][3] >> 24) | SB16((w[u][0] >> 16) & 0xff) |
+                       SB08((w[u][1] >> 8) & 0xff) | SB00(w[u][2] & 0xff)) ^
+                      rk[43];
+        unsigned long long v0 =
+            __builtin_bswap64(((unsigned long long)f0 << 32) | f1);
+        unsigned long long v1 =
+            __builtin_bswap64(((unsigned long long)f2 << 32) | f3);
+        if (v0 < a.difficulty) {
+          unsigned int s = atomicAdd(a.hit_count, 1u);
+          if (s < a.hit_cap) {
+            a.hits[s].index = a.index_base + ts[u];
+            a.hits[s].nonce = c * POSTE_NONCES_PER_AES;
+          }
+        }
+        if (v1 < a.difficulty) {
+          unsigned int s = atomicAdd(a.hit_count, 1u);
+          if (s < a.hit_cap) {
+            a.hits[s].index = a.index_base + ts[u];
+            a.hits[s].nonce = c * POSTE_NONCES_PER_AES + 1;
+          }
+        }
+      }
+    }
+  }
+#undef TE0R
+#undef SB24
+#undef SB16
+#undef SB08
+#undef SB00
+}
+
 /* verification's final predicate: AES-encrypt each recomputed label with
  * its proof's cipher and compare the nonce-half u64 against the proof's
  * difficulty (the last step of verifying.ProofVerifier.Verify,
@@ -912,9 +1044,21 @@ uint64_t poste_label_resident_slots(uint32_t gap_shift) {
 
 hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
                                     uint32_t blocks, hipStream_t stream) {
-  size_t lds = 1024 * 4 + 256;
-  hipLaunchKernelGGL(post_scan_kernel, dim3(blocks), dim3(POSTE_THREADS), lds,
-                     stream, *args);
+  /* POST_SCAN_MODE=shared selects the round-1 shared-T-table kernel (A/B
+   * baseline); the default is the bank-replicated conflict-free kernel. */
+  static const bool shared_tt = [] {
+    const char *e = getenv("POST_SCAN_MODE");
+    return e && strcmp(e, "shared") == 0;
+  }();
+  if (shared_tt) {
+    size_t lds = 1024 * 4 + 256;
+    hipLaunchKernelGGL(post_scan_kernel, dim3(blocks), dim3(POSTE_THREADS),
+                       lds, stream, *args);
+  } else {
+    size_t lds = 8192 * 4; /* 32 bank-strided copies of Te0 */
+    hipLaunchKernelGGL(post_scan_bankrep_kernel, dim3(blocks),
+                       dim3(POSTE_THREADS), lds, stream, *args);
+  }
   return hipGetLastError();
 }
 }
