@@ -836,7 +836,13 @@ post_scan_bankrep_kernel(ScanKernelArgs a) {
                          ? (int)((a.count - t0 + stride - 1) / stride)
                          : L;
     for (uint32_t c = 0; c < a.n_ciphers; c++) {
-      const uint32_t *rk = a.rk + c * 44;
+      /* expanded key into registers up front: a.rk accesses inside the
+       * round loop compile to per-round VECTOR global loads (wave-uniform
+       * but the compiler cannot prove it) whose vmcnt waits serialize the
+       * cipher loop */
+      uint32_t rk[44];
+#pragma unroll
+      for (int k = 0; k < 44; k++) rk[k] = a.rk[c * 44 + k];
       uint32_t w[L][4];
 #pragma unroll
       for (int u = 0; u < L; u++)
@@ -912,6 +918,128 @@ post_scan_bankrep_kernel(ScanKernelArgs a) {
 #undef SB16
 #undef SB08
 #undef SB00
+}
+
+/* 4-table interleaved bank-replicated scan: all four Te tables, each
+ * bank-replicated, interleaved so entry (x, t, lane) sits at word
+ * x*128 + t*32 + lane%32 — per-gather addressing is one v_lshl_add with
+ * the t*128-byte offset folded into the ds_read immediate, and the
+ * alignbit rotations of the 1-table variant disappear (~20% of its
+ * VALU).  Costs 128 KiB LDS, so one 1024-thread workgroup per CU
+ * (16 waves) instead of 20. */
+__global__ void __launch_bounds__(1024)
+post_scan_tt4_kernel(ScanKernelArgs a) {
+  extern __shared__ uint32_t sTe[]; /* 32768 words = 4 tables x 32 copies */
+  for (uint32_t i = threadIdx.x; i < 32768; i += blockDim.x)
+    sTe[i] = a.te[((i >> 5) & 3) * 256 + (i >> 7)];
+  __syncthreads();
+  const uint32_t *t0p = sTe + (threadIdx.x & 31);
+#define TT4(t, idx) t0p[(((uint32_t)(idx)) << 7) + 32 * (t)]
+
+  const unsigned long long stride =
+      (unsigned long long)gridDim.x * blockDim.x;
+  constexpr int L = POSTE_SCAN_ILP;
+  const unsigned long long span = stride * L;
+  for (unsigned long long t0 =
+           (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+       t0 < a.count; t0 += span) {
+    uint32_t p[L][4];
+    unsigned long long ts[L];
+#pragma unroll
+    for (int u = 0; u < L; u++) {
+      unsigned long long t = t0 + (unsigned long long)u * stride;
+      ts[u] = t < a.count ? t : t0;
+      uint4 lraw = a.labels[ts[u]];
+      p[u][0] = __builtin_bswap32(lraw.x);
+      p[u][1] = __builtin_bswap32(lraw.y);
+      p[u][2] = __builtin_bswap32(lraw.z);
+      p[u][3] = __builtin_bswap32(lraw.w);
+    }
+    const int live = (int)((a.count - t0 + stride - 1) / stride) < L
+                         ? (int)((a.count - t0 + stride - 1) / stride)
+                         : L;
+    for (uint32_t c = 0; c < a.n_ciphers; c++) {
+      uint32_t rk[44];
+#pragma unroll
+      for (int k = 0; k < 44; k++) rk[k] = a.rk[c * 44 + k];
+      uint32_t w[L][4];
+#pragma unroll
+      for (int u = 0; u < L; u++)
+#pragma unroll
+        for (int k = 0; k < 4; k++) w[u][k] = p[u][k] ^ rk[k];
+#pragma unroll
+      for (int r = 1; r < 10; r++) {
+#pragma unroll
+        for (int u = 0; u < L; u++) {
+          uint32_t n0 = TT4(0, w[u][0] >> 24) ^
+                        TT4(1, (w[u][1] >> 16) & 0xff) ^
+                        TT4(2, (w[u][2] >> 8) & 0xff) ^
+                        TT4(3, w[u][3] & 0xff) ^ rk[4 * r];
+          uint32_t n1 = TT4(0, w[u][1] >> 24) ^
+                        TT4(1, (w[u][2] >> 16) & 0xff) ^
+                        TT4(2, (w[u][3] >> 8) & 0xff) ^
+                        TT4(3, w[u][0] & 0xff) ^ rk[4 * r + 1];
+          uint32_t n2 = TT4(0, w[u][2] >> 24) ^
+                        TT4(1, (w[u][3] >> 16) & 0xff) ^
+                        TT4(2, (w[u][0] >> 8) & 0xff) ^
+                        TT4(3, w[u][1] & 0xff) ^ rk[4 * r + 2];
+          uint32_t n3 = TT4(0, w[u][3] >> 24) ^
+                        TT4(1, (w[u][0] >> 16) & 0xff) ^
+                        TT4(2, (w[u][1] >> 8) & 0xff) ^
+                        TT4(3, w[u][2] & 0xff) ^ rk[4 * r + 3];
+          w[u][0] = n0; w[u][1] = n1; w[u][2] = n2; w[u][3] = n3;
+        }
+      }
+      /* final round: S[x] = (Te0[x]>>8)&0xff, t=0 plane */
+#define SB24_4(idx) ((TT4(0, idx) & 0xff00u) << 16)
+#define SB16_4(idx) ((TT4(0, idx) & 0xff00u) << 8)
+#define SB08_4(idx) (TT4(0, idx) & 0xff00u)
+#define SB00_4(idx) ((TT4(0, idx) >> 8) & 0xffu)
+#pragma unroll
+      for (int u = 0; u < L; u++) {
+        if (u >= live) break;
+        uint32_t f0 = (SB24_4(w[u][0] >> 24) |
+                       SB16_4((w[u][1] >> 16) & 0xff) |
+                       SB08_4((w[u][2] >> 8) & 0xff) |
+                       SB00_4(w[u][3] & 0xff)) ^ rk[40];
+        uint32_t f1 = (SB24_4(w[u][1] >> 24) |
+                       SB16_4((w[u][2] >> 16) & 0xff) |
+                       SB08_4((w[u][3] >> 8) & 0xff) |
+                       SB00_4(w[u][0] & 0xff)) ^ rk[41];
+        uint32_t f2 = (SB24_4(w[u][2] >> 24) |
+                       SB16_4((w[u][3] >> 16) & 0xff) |
+                       SB08_4((w[u][0] >> 8) & 0xff) |
+                       SB00_4(w[u][1] & 0xff)) ^ rk[42];
+        uint32_t f3 = (SB24_4(w[u][3] >> 24) |
+                       SB16_4((w[u][0] >> 16) & 0xff) |
+                       SB08_4((w[u][1] >> 8) & 0xff) |
+                       SB00_4(w[u][2] & 0xff)) ^ rk[43];
+        unsigned long long v0 =
+            __builtin_bswap64(((unsigned long long)f0 << 32) | f1);
+        unsigned long long v1 =
+            __builtin_bswap64(((unsigned long long)f2 << 32) | f3);
+        if (v0 < a.difficulty) {
+          unsigned int s = atomicAdd(a.hit_count, 1u);
+          if (s < a.hit_cap) {
+            a.hits[s].index = a.index_base + ts[u];
+            a.hits[s].nonce = c * POSTE_NONCES_PER_AES;
+          }
+        }
+        if (v1 < a.difficulty) {
+          unsigned int s = atomicAdd(a.hit_count, 1u);
+          if (s < a.hit_cap) {
+            a.hits[s].index = a.index_base + ts[u];
+            a.hits[s].nonce = c * POSTE_NONCES_PER_AES + 1;
+          }
+        }
+      }
+    }
+  }
+#undef TT4
+#undef SB24_4
+#undef SB16_4
+#undef SB08_4
+#undef SB00_4
 }
 
 /* verification's final predicate: AES-encrypt each recomputed label with
@@ -1044,16 +1172,37 @@ uint64_t poste_label_resident_slots(uint32_t gap_shift) {
 
 hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
                                     uint32_t blocks, hipStream_t stream) {
-  /* POST_SCAN_MODE=shared selects the round-1 shared-T-table kernel (A/B
-   * baseline); the default is the bank-replicated conflict-free kernel. */
-  static const bool shared_tt = [] {
+  /* POST_SCAN_MODE: shared = round-1 shared-T-table kernel, tt4 =
+   * 4-table/128KiB replicated kernel; default = bank-replicated Te0. */
+  static const int mode = [] {
     const char *e = getenv("POST_SCAN_MODE");
-    return e && strcmp(e, "shared") == 0;
+    if (e && strcmp(e, "shared") == 0) return 0;
+    if (e && strcmp(e, "tt4") == 0) return 2;
+    return 1;
   }();
-  if (shared_tt) {
+  if (mode == 0) {
     size_t lds = 1024 * 4 + 256;
     hipLaunchKernelGGL(post_scan_kernel, dim3(blocks), dim3(POSTE_THREADS),
                        lds, stream, *args);
+  } else if (mode == 2) {
+    static bool attr_ok = [] {
+      return hipFuncSetAttribute(
+                 (const void *)post_scan_tt4_kernel,
+                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                 32768 * 4) == hipSuccess;
+    }();
+    if (attr_ok) {
+      uint32_t b4 = (uint32_t)((args->count + 1023) / 1024);
+      if (b4 > 2048) b4 = 2048;
+      if (b4 == 0) b4 = 1;
+      hipLaunchKernelGGL(post_scan_tt4_kernel, dim3(b4), dim3(1024),
+                         32768 * 4, stream, *args);
+      return hipGetLastError();
+    }
+    /* 128 KiB dynamic LDS unavailable: fall through to bankrep */
+    size_t lds = 8192 * 4;
+    hipLaunchKernelGGL(post_scan_bankrep_kernel, dim3(blocks),
+                       dim3(POSTE_THREADS), lds, stream, *args);
   } else {
     size_t lds = 8192 * 4; /* 32 bank-strided copies of Te0 */
     hipLaunchKernelGGL(post_scan_bankrep_kernel, dim3(blocks),

@@ -449,6 +449,62 @@ def test_concurrent_verify_threads(roundtrip):
     assert results.count("rejected") == 8
 
 
+def test_concurrent_mixed_ops(roundtrip, tmp_path):
+    """Whole-ABI concurrency contract (the strongest form available with
+    no Go toolchain to compile the cgo shim): one thread
+    runs an init session, one proves, N verify, one polls
+    providers/benchmark/selftests — all against the same engine library at
+    once, like a node initializing one identity while gossip-verifying and
+    self-proving another."""
+    from concurrent.futures import ThreadPoolExecutor
+    NU, LPU, N, labels, proof, _ = roundtrip
+    vcfg = gsm_amd.PostConfig(k1=12, k2=8, k3=8, pow_difficulty=POW_DIFF)
+    meta = gsm_amd.PostProofMetadata(NODE, ATX, CHALLENGE, NU, LPU)
+
+    def do_init():
+        cfg, mgr = make_mgr(1, 1 << 12, 128, data_dir=str(tmp_path),
+                            max_file_size=1 << 15)
+        mgr.prepare_initializer()
+        mgr.start_session()
+        st = mgr.status()
+        mgr.reset()
+        return ("init", st["num_labels_written"] == 1 << 12)
+
+    def do_prove():
+        pcfg = gsm_amd.PostConfig(min_num_units=1, labels_per_unit=LPU,
+                                  k1=12, k2=8, k3=4,
+                                  pow_difficulty=POW_DIFF)
+        pr = gsm_amd.api.prove_buffer(labels, NU * LPU, NODE, ATX,
+                                      CHALLENGE, pcfg,
+                                      gsm_amd.ProveOpts(nonces=16))
+        return ("prove", pr.nonce == proof.nonce and
+                pr.indices == proof.indices)
+
+    def do_verify(i):
+        ver = gsm_amd.PostVerifier(vcfg, scrypt_n=N)
+        ver.verify(proof, meta)
+        return ("verify", True)
+
+    def do_poll():
+        eng = gsm_amd.api.Engine()
+        provs = eng.providers()
+        b = eng.benchmark(provider_id=0, scrypt_n=128)
+        h = eng.selftest_blake3(b"concurrency")
+        lab = eng.selftest_label(NODE, ATX, 5, 128)
+        o = Oracle()
+        commit = o.commitment(NODE, ATX)
+        want, _ = o.init_range(commit, 5, 1, 128)
+        return ("poll", len(provs) >= 1 and b > 0 and len(h) == 32 and
+                lab[:16] == want)
+
+    jobs = [do_init, do_prove, do_poll] + [lambda i=i: do_verify(i)
+                                           for i in range(5)]
+    with ThreadPoolExecutor(8) as ex:
+        futs = [ex.submit(j) for j in jobs]
+        results = [f.result(timeout=300) for f in futs]
+    assert all(ok for _, ok in results), results
+
+
 def test_cfg1_checksum_regression():
     """Determinism pin across kernel changes: sha256 over the first 2^20
     labels of BASELINE config 1 (mainnet N, fixed identity).  The absolute
