@@ -800,7 +800,8 @@ post_scan_kernel(ScanKernelArgs a) {
 #define ROR16(x) __builtin_amdgcn_alignbit((x), (x), 16)
 #define ROR24(x) __builtin_amdgcn_alignbit((x), (x), 24)
 
-__global__ void __launch_bounds__(POSTE_THREADS)
+template <int THREADS_, bool BATCHG>
+__global__ void __launch_bounds__(THREADS_)
 post_scan_bankrep_kernel(ScanKernelArgs a) {
   extern __shared__ uint32_t sTe[]; /* 8192 words = 32 copies x 1 KiB */
   for (uint32_t i = threadIdx.x; i < 8192; i += blockDim.x)
@@ -850,6 +851,28 @@ post_scan_bankrep_kernel(ScanKernelArgs a) {
         for (int k = 0; k < 4; k++) w[u][k] = p[u][k] ^ rk[k];
 #pragma unroll
       for (int r = 1; r < 10; r++) {
+        if (BATCHG) {
+          /* issue all of the round's gathers before any combine: widens
+           * the read->use distance so partial lgkmcnt waits overlap the
+           * whole gather batch with the previous combines */
+          uint32_t g[L][16];
+#pragma unroll
+          for (int u = 0; u < L; u++)
+#pragma unroll
+            for (int q = 0; q < 4; q++) {
+              g[u][4 * q + 0] = TE0R(w[u][q] >> 24);
+              g[u][4 * q + 1] = TE0R((w[u][(q + 1) & 3] >> 16) & 0xff);
+              g[u][4 * q + 2] = TE0R((w[u][(q + 2) & 3] >> 8) & 0xff);
+              g[u][4 * q + 3] = TE0R(w[u][(q + 3) & 3] & 0xff);
+            }
+#pragma unroll
+          for (int u = 0; u < L; u++)
+#pragma unroll
+            for (int q = 0; q < 4; q++)
+              w[u][q] = g[u][4 * q] ^ ROR8(g[u][4 * q + 1]) ^
+                        ROR16(g[u][4 * q + 2]) ^ ROR24(g[u][4 * q + 3]) ^
+                        rk[4 * r + q];
+        } else {
 #pragma unroll
         for (int u = 0; u < L; u++) {
           uint32_t n0 = TE0R(w[u][0] >> 24) ^
@@ -869,6 +892,7 @@ post_scan_bankrep_kernel(ScanKernelArgs a) {
                         ROR16(TE0R((w[u][1] >> 8) & 0xff)) ^
                         ROR24(TE0R(w[u][2] & 0xff)) ^ rk[4 * r + 3];
           w[u][0] = n0; w[u][1] = n1; w[u][2] = n2; w[u][3] = n3;
+        }
         }
       }
       /* final round: S[x] = (Te0[x]>>8)&0xff; assemble each output word
@@ -1172,19 +1196,21 @@ uint64_t poste_label_resident_slots(uint32_t gap_shift) {
 
 hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
                                     uint32_t blocks, hipStream_t stream) {
-  /* POST_SCAN_MODE: shared = round-1 shared-T-table kernel, tt4 =
-   * 4-table/128KiB replicated kernel; default = bank-replicated Te0. */
-  static const int mode = [] {
-    const char *e = getenv("POST_SCAN_MODE");
-    if (e && strcmp(e, "shared") == 0) return 0;
-    if (e && strcmp(e, "tt4") == 0) return 2;
-    return 1;
-  }();
-  if (mode == 0) {
+  /* POST_SCAN_MODE: shared = round-1 shared-T-table kernel; tt4 =
+   * 4-table/128KiB replicated kernel; bankrep512 / bankrepbg /
+   * bankrep512bg = bank-replicated at 512-thread workgroups and/or
+   * batched-gather scheduling; default = bank-replicated Te0 at 256.
+   * Read per launch (launches are ms-scale) so tests can A/B kernels
+   * within one process. */
+  const char *e = getenv("POST_SCAN_MODE");
+  const char *m = e ? e : "bankrep";
+  if (strcmp(m, "shared") == 0) {
     size_t lds = 1024 * 4 + 256;
     hipLaunchKernelGGL(post_scan_kernel, dim3(blocks), dim3(POSTE_THREADS),
                        lds, stream, *args);
-  } else if (mode == 2) {
+    return hipGetLastError();
+  }
+  if (strcmp(m, "tt4") == 0) {
     static bool attr_ok = [] {
       return hipFuncSetAttribute(
                  (const void *)post_scan_tt4_kernel,
@@ -1199,14 +1225,29 @@ hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
                          32768 * 4, stream, *args);
       return hipGetLastError();
     }
-    /* 128 KiB dynamic LDS unavailable: fall through to bankrep */
-    size_t lds = 8192 * 4;
-    hipLaunchKernelGGL(post_scan_bankrep_kernel, dim3(blocks),
-                       dim3(POSTE_THREADS), lds, stream, *args);
+    m = "bankrep"; /* 128 KiB dynamic LDS unavailable */
+  }
+  const bool bg = strstr(m, "bg") != nullptr;
+  const bool wide = strstr(m, "512") != nullptr;
+  size_t lds = 8192 * 4; /* 32 bank-strided copies of Te0 */
+  if (wide) {
+    uint32_t b5 = (uint32_t)((args->count + 511) / 512);
+    if (b5 > 4096) b5 = 4096;
+    if (b5 == 0) b5 = 1;
+    if (bg)
+      hipLaunchKernelGGL((post_scan_bankrep_kernel<512, true>), dim3(b5),
+                         dim3(512), lds, stream, *args);
+    else
+      hipLaunchKernelGGL((post_scan_bankrep_kernel<512, false>), dim3(b5),
+                         dim3(512), lds, stream, *args);
+  } else if (bg) {
+    hipLaunchKernelGGL((post_scan_bankrep_kernel<POSTE_THREADS, true>),
+                       dim3(blocks), dim3(POSTE_THREADS), lds, stream,
+                       *args);
   } else {
-    size_t lds = 8192 * 4; /* 32 bank-strided copies of Te0 */
-    hipLaunchKernelGGL(post_scan_bankrep_kernel, dim3(blocks),
-                       dim3(POSTE_THREADS), lds, stream, *args);
+    hipLaunchKernelGGL((post_scan_bankrep_kernel<POSTE_THREADS, false>),
+                       dim3(blocks), dim3(POSTE_THREADS), lds, stream,
+                       *args);
   }
   return hipGetLastError();
 }

@@ -505,6 +505,27 @@ def test_concurrent_mixed_ops(roundtrip, tmp_path):
     assert all(ok for _, ok in results), results
 
 
+def test_scan_kernel_variants_agree(roundtrip):
+    """All three scan kernels (shared T-table, bank-replicated Te0,
+    4-table interleaved) must produce the identical proof on the same
+    labels — they implement one algorithm with different LDS layouts."""
+    NU, LPU, N, labels, proof, _ = roundtrip
+    cfg = gsm_amd.PostConfig(k1=12, k2=8, k3=4, pow_difficulty=POW_DIFF)
+    results = {}
+    for mode in ["shared", "bankrep", "tt4"]:
+        os.environ["POST_SCAN_MODE"] = mode
+        try:
+            pr = gsm_amd.api.prove_buffer(labels, NU * LPU, NODE, ATX,
+                                          CHALLENGE, cfg,
+                                          gsm_amd.ProveOpts(nonces=16))
+            results[mode] = (pr.nonce, pr.indices, pr.pow)
+        finally:
+            del os.environ["POST_SCAN_MODE"]
+    assert results["shared"] == (proof.nonce, proof.indices, proof.pow)
+    assert results["bankrep"] == results["shared"]
+    assert results["tt4"] == results["shared"]
+
+
 def test_cfg1_checksum_regression():
     """Determinism pin across kernel changes: sha256 over the first 2^20
     labels of BASELINE config 1 (mainnet N, fixed identity).  The absolute
