@@ -962,7 +962,9 @@ post_scan_tt4_kernel(ScanKernelArgs a) {
 
   const unsigned long long stride =
       (unsigned long long)gridDim.x * blockDim.x;
-  constexpr int L = POSTE_SCAN_ILP;
+  /* ILP 4 measured best for this variant (r2d sweep); at one workgroup
+   * per CU there are only 16 waves to hide latency, so deeper chains pay */
+  constexpr int L = POSTE_SCAN_ILP > 4 ? POSTE_SCAN_ILP : 4;
   const unsigned long long span = stride * L;
   for (unsigned long long t0 =
            (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1203,7 +1205,8 @@ hipError_t poste_launch_scan_kernel(const ScanKernelArgs *args,
    * Read per launch (launches are ms-scale) so tests can A/B kernels
    * within one process. */
   const char *e = getenv("POST_SCAN_MODE");
-  const char *m = e ? e : "bankrep";
+  const char *m = e ? e : "tt4"; /* fastest measured (r2d sweep); falls
+                                    back to bankrep without 128 KiB LDS */
   if (strcmp(m, "shared") == 0) {
     size_t lds = 1024 * 4 + 256;
     hipLaunchKernelGGL(post_scan_kernel, dim3(blocks), dim3(POSTE_THREADS),
