@@ -23,6 +23,7 @@ import time
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
+sys.path.insert(0, os.path.join(REPO, "oracle"))  # oracle/oracle.py as module
 
 import gsm_amd  # noqa: E402
 from oracle import Oracle  # noqa: E402  (checker only: parity spot-checks)
