@@ -701,6 +701,7 @@ int post_init_step(PostInitSession *s, uint64_t max_labels, uint64_t *done) {
                                          s->cfg.scrypt_n, ref)
                      : 0;
     HIP_TRY(hipStreamSynchronize(s->stream));
+    bool nonce_improved = false;
     if (n_cand > 0) {
       unsigned int take = std::min(n_cand, CAND_CAP);
       HIP_TRY(hipMemcpy(cands.data(), s->d_cand,
@@ -723,8 +724,19 @@ int post_init_step(PostInitSession *s, uint64_t max_labels, uint64_t *done) {
           s->nonce_idx = cands[i].index;
           std::memcpy(s->nonce_label, lab, 32);
           std::memcpy(cur_difficulty, lab, 32);
+          nonce_improved = true;
         }
       }
+    }
+    /* persist an improved running minimum immediately: metadata is the
+     * only carrier of the shard-local minimum across kill+resume (the
+     * creation-time write has no nonce yet, and a completed-at-exit-only
+     * write loses it on any interrupt — the cfg4 soak caught exactly
+     * that).  Improvements are rare (~log of batches), so this is a few
+     * small rewrites per session. */
+    if (nonce_improved && !s->data_dir.empty()) {
+      rc = write_metadata(s);
+      if (rc != POST_OK) return rc;
     }
 
     /* reference-label self-check (ErrReferenceLabelMismatch semantics,
@@ -799,22 +811,30 @@ static int nonce_only_batch(PostInitSession *s, uint64_t gstart,
     HIP_TRY(hipMemcpy(cands.data(), s->d_cand,
                       sizeof(PostVrfCandidate) * take,
                       hipMemcpyDeviceToHost));
-    std::lock_guard<std::mutex> lk(s->nonce_mu);
-    for (unsigned int i = 0; i < take; i++) {
-      uint8_t lab[32];
-      for (int k = 0; k < 8; k++) {
-        uint32_t w = cands[i].label_be[k];
-        lab[4 * k] = (uint8_t)(w >> 24);
-        lab[4 * k + 1] = (uint8_t)(w >> 16);
-        lab[4 * k + 2] = (uint8_t)(w >> 8);
-        lab[4 * k + 3] = (uint8_t)w;
+    bool improved = false;
+    {
+      std::lock_guard<std::mutex> lk(s->nonce_mu);
+      for (unsigned int i = 0; i < take; i++) {
+        uint8_t lab[32];
+        for (int k = 0; k < 8; k++) {
+          uint32_t w = cands[i].label_be[k];
+          lab[4 * k] = (uint8_t)(w >> 24);
+          lab[4 * k + 1] = (uint8_t)(w >> 16);
+          lab[4 * k + 2] = (uint8_t)(w >> 8);
+          lab[4 * k + 3] = (uint8_t)w;
+        }
+        int c = s->nonce_found ? std::memcmp(lab, s->nonce_label, 32) : -1;
+        if (c < 0 || (c == 0 && cands[i].index < s->nonce_idx)) {
+          s->nonce_found = true;
+          s->nonce_idx = cands[i].index;
+          std::memcpy(s->nonce_label, lab, 32);
+          improved = true;
+        }
       }
-      int c = s->nonce_found ? std::memcmp(lab, s->nonce_label, 32) : -1;
-      if (c < 0 || (c == 0 && cands[i].index < s->nonce_idx)) {
-        s->nonce_found = true;
-        s->nonce_idx = cands[i].index;
-        std::memcpy(s->nonce_label, lab, 32);
-      }
+    }
+    if (improved && !s->data_dir.empty()) {
+      int rc = write_metadata(s);
+      if (rc != POST_OK) return rc;
     }
   }
   return POST_OK;

@@ -493,7 +493,7 @@ def test_concurrent_mixed_ops(roundtrip, tmp_path):
         lab = eng.selftest_label(NODE, ATX, 5, 128)
         o = Oracle()
         commit = o.commitment(NODE, ATX)
-        want, _ = o.init_range(commit, 5, 1, 128)
+        want, _ = o.init_range(commit, 5, 6, 128)
         return ("poll", len(provs) >= 1 and b > 0 and len(h) == 32 and
                 lab[:16] == want)
 
