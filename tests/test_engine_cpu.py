@@ -131,3 +131,23 @@ def test_num_units_bounds_enforced():
         with pytest.raises(gsm_amd.EngineError) as ei:
             mgr.prepare_initializer()
         assert ei.value.code == gsm_amd.api.Status.INVALID_ARGS
+
+
+def test_roofline_traffic_record_contract():
+    """bench.py's roofline.traffic rides from the committed PMC record
+    (profiles/roofline_traffic.json); guard the fields it reads so a
+    kernel change that forgets to re-measure fails fast here."""
+    import json
+    import os
+    rec = json.load(open(os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "profiles", "roofline_traffic.json")))
+    assert rec["kernel"] == "post_label_romix_kernel"
+    assert rec["labels_per_launch"] > 0
+    assert rec["traffic_bytes_per_launch"] == (
+        rec["write_bytes_per_launch"] + rec["fetch_bytes_per_launch"])
+    # counter-measured traffic must stay within ~5% of algorithmic bytes
+    # (128*N*2 + 512 + 16 per label at N=8192) — "no wasted re-reads"
+    algo = (128 * 8192 * 2 + 512 + 16) * rec["labels_per_launch"]
+    assert abs(rec["traffic_bytes_per_launch"] - algo) / algo < 0.05
+    assert "source" in rec and rec["source"].startswith("profiles/")

@@ -26,4 +26,14 @@ echo "== 4. aux benches (warmed: scan e2e + verify batched/single)"
 timeout -k 10 600 python bench_aux.py --scan-labels 24 --verify-proofs 10000 \
   > "$OUT/r2_aux.json" 2>&1
 tail -6 "$OUT/r2_aux.json"
+
+echo "== 5. 2-rank bench dry run on one GPU (SCALE-path validation: rank"
+echo "      sharding + nonce min-reduce + two engine sessions on one device)"
+POST_BENCH_BACKEND=gloo POST_BENCH_SCRATCH=$((60 * 1024 * 1024 * 1024)) \
+  POST_SKIP_CPU_BASELINE=1 \
+  timeout -k 10 420 python -m torch.distributed.run --nnodes=1 \
+  --nproc-per-node 2 --master-addr 127.0.0.1 --master-port 29317 \
+  bench.py --gpus 2 --steps 2 --warmup 1 > "$OUT/r2_bench2rank.json" 2>&1
+echo "2rank rc=$?"
+grep -o '"metric[^}]*' "$OUT/r2_bench2rank.json" | head -1 || tail -5 "$OUT/r2_bench2rank.json"
 echo done
