@@ -687,11 +687,9 @@ int post_init_step(PostInitSession *s, uint64_t max_labels, uint64_t *done) {
      * scrypt otherwise contends for the same host cores (SCALE readiness,
      * VERDICT r01 §next-3); detection latency rises k batches, coverage
      * semantics (ErrReferenceLabelMismatch) are unchanged. */
-    static const long selfcheck_period = [] {
-      const char *e = getenv("POST_SELFCHECK_PERIOD");
-      long v = e ? strtol(e, nullptr, 10) : 1;
-      return v < 1 ? 1L : v;
-    }();
+    const char *sp_env = getenv("POST_SELFCHECK_PERIOD");
+    long selfcheck_period = sp_env ? strtol(sp_env, nullptr, 10) : 1;
+    if (selfcheck_period < 1) selfcheck_period = 1;
     const bool do_selfcheck = (s->batch_seq++ % (uint64_t)selfcheck_period)
                               == 0;
     const uint64_t probe = count / 2;

@@ -505,6 +505,24 @@ def test_concurrent_mixed_ops(roundtrip, tmp_path):
     assert all(ok for _, ok in results), results
 
 
+def test_selfcheck_period_preserves_parity():
+    """POST_SELFCHECK_PERIOD thins the host reference-label check (multi-
+    rank contention knob) but must not change any output byte."""
+    os.environ["POST_SELFCHECK_PERIOD"] = "4"
+    try:
+        cfg, mgr = make_mgr(1, 1 << 13, 128)
+        mgr.prepare_initializer()
+        mgr.start_session()
+        got = mgr.copy_labels(0, 1 << 13)
+        mgr.reset()
+    finally:
+        del os.environ["POST_SELFCHECK_PERIOD"]
+    o = Oracle()
+    commit = o.commitment(NODE, ATX)
+    want, _ = o.init_range(commit, 0, 1 << 13, 128)
+    assert got == want
+
+
 def test_scan_kernel_variants_agree(roundtrip):
     """All three scan kernels (shared T-table, bank-replicated Te0,
     4-table interleaved) must produce the identical proof on the same
