@@ -36,4 +36,14 @@ POST_BENCH_BACKEND=gloo POST_BENCH_SCRATCH=$((60 * 1024 * 1024 * 1024)) \
   bench.py --gpus 2 --steps 2 --warmup 1 > "$OUT/r2_bench2rank.json" 2>&1
 echo "2rank rc=$?"
 grep -o '"metric[^}]*' "$OUT/r2_bench2rank.json" | head -1 || tail -5 "$OUT/r2_bench2rank.json"
+
+echo "== 6. 8-rank dry run on one GPU (the driver's N=8 torchrun shape;"
+echo "      8 engine sessions share the device at bounded scratch)"
+POST_BENCH_BACKEND=gloo POST_BENCH_SCRATCH=$((16 * 1024 * 1024 * 1024)) \
+  POST_SKIP_CPU_BASELINE=1 \
+  timeout -k 10 420 python -m torch.distributed.run --nnodes=1 \
+  --nproc-per-node 8 --master-addr 127.0.0.1 --master-port 29319 \
+  bench.py --gpus 8 --steps 1 --warmup 1 > "$OUT/r2_bench8rank.json" 2>&1
+echo "8rank rc=$?"
+grep -o '"metric[^}]*' "$OUT/r2_bench8rank.json" | head -1 || tail -5 "$OUT/r2_bench8rank.json"
 echo done
