@@ -20,6 +20,7 @@
 #include <algorithm>
 #include <array>
 #include <atomic>
+#include <thread>
 #include <chrono>
 #include <cstdio>
 #include <cstdlib>
@@ -1320,39 +1321,41 @@ int post_verify_batch_seeded(const PostProof *proofs,
     uint32_t position; /* position within the proof's k2 indices */
   };
   std::vector<Task> tasks;
-  std::vector<uint32_t> commit_words; /* 8 words per proof */
+  std::vector<uint32_t> commit_words((size_t)n * 8);
   std::vector<std::vector<uint64_t>> proof_indices(n);
+  std::vector<std::vector<Task>> per_proof_tasks(n);
 
-  for (uint32_t p = 0; p < n; p++) {
+  /* per-proof host prep (commitment blake3, pow verify, index unpack,
+   * subset sampling) is independent across proofs and dominates the
+   * K3=1 batch wall time — spread it across host threads; every slot
+   * written is per-proof disjoint.  The commitment table is indexed by
+   * ABSOLUTE proof number from the kernel tasks, so every proof gets a
+   * slot — including ones rejected here. */
+  auto prep_one = [&](uint32_t p) {
     statuses[p] = POST_OK;
     const PostProof &pr = proofs[p];
     const PostProofMetadata &me = metas[p];
-    /* the commitment table is indexed by ABSOLUTE proof number from the
-     * kernel tasks, so every proof gets a slot — including ones rejected
-     * below */
     {
       uint8_t cm[32];
       poste::commitment(me.node_id, me.commitment_atx_id, cm);
-      uint32_t cw[8];
-      std::memcpy(cw, cm, 32);
-      for (int k = 0; k < 8; k++) commit_words.push_back(cw[k]);
+      std::memcpy(commit_words.data() + (size_t)p * 8, cm, 32);
     }
     uint64_t num_labels = (uint64_t)me.num_units * me.labels_per_unit;
     if (num_labels == 0) { /* malformed metadata */
       statuses[p] = POST_ERR_INVALID_ARGS;
-      continue;
+      return;
     }
     uint32_t bpi = poste::bits_per_index(num_labels);
     if (pr.num_indices != cfg->k2 ||
         pr.indices_len != ((uint64_t)cfg->k2 * bpi + 7) / 8) {
       statuses[p] = POST_ERR_INVALID_ARGS;
-      continue;
+      return;
     }
     uint32_t group = pr.nonce / POSTE_NONCE_GROUP;
     if (poste::k2pow_verify_blake3(me.challenge, group, pr.pow,
                                    cfg->pow_difficulty) != 0) {
       statuses[p] = POST_ERR_POW;
-      continue;
+      return;
     }
     proof_indices[p].resize(cfg->k2);
     poste::unpack_indices(pr.indices, cfg->k2, bpi,
@@ -1375,24 +1378,44 @@ int post_verify_batch_seeded(const PostProof *proofs,
     for (uint32_t pos : positions) {
       if (pos >= cfg->k2) {
         statuses[p] = POST_ERR_INVALID_ARGS;
-        break;
+        return;
       }
       uint64_t li = proof_indices[p][pos];
       if (li >= num_labels) {
         statuses[p] = POST_ERR_INVALID_INDEX;
         if (invalid_indices) invalid_indices[p] = pos;
-        break;
+        return;
       }
-      tasks.push_back({li, p, pos});
+      per_proof_tasks[p].push_back({li, p, pos});
+    }
+  };
+  {
+    unsigned hw = std::thread::hardware_concurrency();
+    unsigned nthreads = std::min<unsigned>(hw ? hw : 1, 32);
+    if (n < 64) nthreads = 1; /* not worth spawning for small batches */
+    if (nthreads <= 1) {
+      for (uint32_t p = 0; p < n; p++) prep_one(p);
+    } else {
+      std::vector<std::thread> ths;
+      std::atomic<uint32_t> next{0};
+      for (unsigned t = 0; t < nthreads; t++)
+        ths.emplace_back([&] {
+          for (;;) {
+            uint32_t p = next.fetch_add(64);
+            if (p >= n) return;
+            uint32_t e = std::min(n, p + 64);
+            for (; p < e; p++) prep_one(p);
+          }
+        });
+      for (auto &th : ths) th.join();
     }
   }
-
-  /* drop tasks of proofs already failed */
-  tasks.erase(std::remove_if(tasks.begin(), tasks.end(),
-                             [&](const Task &t) {
-                               return statuses[t.proof] != POST_OK;
-                             }),
-              tasks.end());
+  /* concatenate in proof order (deterministic task->index mapping) */
+  for (uint32_t p = 0; p < n; p++) {
+    if (statuses[p] != POST_OK) continue;
+    tasks.insert(tasks.end(), per_proof_tasks[p].begin(),
+                 per_proof_tasks[p].end());
+  }
   if (tasks.empty()) return POST_OK;
 
   /* scratch-bounded chunks of label recomputes on the GPU */
@@ -1420,22 +1443,26 @@ int post_verify_batch_seeded(const PostProof *proofs,
       max_lanes, ((tasks.size() + 127) / 128) * 128);
 
   /* per-proof cipher data for the device predicate (absolute indexing:
-   * every proof gets a slot, matching task_proof) */
+   * every proof gets a slot, matching task_proof).  Declared here,
+   * COMPUTED after the label kernel launches — the ~10k blake3+AES key
+   * expansions overlap the label recompute instead of preceding it. */
   std::vector<uint32_t> vrk((size_t)n * 44, 0);
   std::vector<uint8_t> vhalf(n, 0);
   std::vector<uint64_t> vdiff(n, 0);
-  for (uint32_t p = 0; p < n; p++) {
-    if (statuses[p] != POST_OK) continue;
-    const PostProof &pr = proofs[p];
-    const PostProofMetadata &me = metas[p];
-    uint8_t key[16];
-    poste::prove_cipher_key(me.challenge, pr.nonce / POSTE_NONCES_PER_AES,
-                            pr.pow, key);
-    poste::aes128_expand(key, vrk.data() + (size_t)p * 44);
-    vhalf[p] = (uint8_t)(pr.nonce % POSTE_NONCES_PER_AES);
-    uint64_t num_labels = (uint64_t)me.num_units * me.labels_per_unit;
-    vdiff[p] = poste::proving_difficulty(cfg->k1, num_labels);
-  }
+  auto cipher_prep = [&] {
+    for (uint32_t p = 0; p < n; p++) {
+      if (statuses[p] != POST_OK) continue;
+      const PostProof &pr = proofs[p];
+      const PostProofMetadata &me = metas[p];
+      uint8_t key[16];
+      poste::prove_cipher_key(me.challenge, pr.nonce / POSTE_NONCES_PER_AES,
+                              pr.pow, key);
+      poste::aes128_expand(key, vrk.data() + (size_t)p * 44);
+      vhalf[p] = (uint8_t)(pr.nonce % POSTE_NONCES_PER_AES);
+      uint64_t num_labels = (uint64_t)me.num_units * me.labels_per_unit;
+      vdiff[p] = poste::proving_difficulty(cfg->k1, num_labels);
+    }
+  };
   DeviceTables tbl;
   rc = get_aes_tables((int)cfg->provider_id, tbl);
   if (rc != POST_OK) return rc;
@@ -1485,6 +1512,7 @@ int post_verify_batch_seeded(const PostProof *proofs,
     }
     HIP_TRY(poste_launch_label_kernel(&la, (uint32_t)(lanes / 64),
                                       nullptr));
+    cipher_prep(); /* overlaps the label kernel (launch is async) */
     if (dbg) {
       (void)hipDeviceSynchronize();
       std::fprintf(stderr, "[verify] label kernel %.3fs\n", tick() - t0);
