@@ -498,19 +498,29 @@ class PostVerifier:
                                   f"{inv.value}")
         self.engine._check(rc)
 
-    def verify_batch(self, proofs, metas, opts: VerifyOpts = VerifyOpts(),
-                     seeds=None):
-        # seeds (optional): equal-length per-proof subset seeds — each
-        # gossip verify samples with its own peer seed
-        # (validation.go:206-209).
+    def marshal_batch(self, proofs, metas):
+        """Pre-build the C argument arrays for verify_batch.  A cgo shim
+        holds these layouts natively; the ctypes conversion is a
+        Python-mirror cost only, so callers that re-verify (or benchmark
+        the ABI boundary) can marshal once and pass the result to
+        verify_batch as `marshalled=`."""
         n = len(proofs)
-        vc, _seed = self._vc(opts)
         cps = (CProof * n)()
         cms = (CProofMetadata * n)()
         for i, (p, m) in enumerate(zip(proofs, metas)):
             cps[i] = p.to_c()
             cps[i].num_indices = self.cfg.k2
             cms[i] = m.to_c()
+        return cps, cms, n
+
+    def verify_batch(self, proofs, metas, opts: VerifyOpts = VerifyOpts(),
+                     seeds=None, marshalled=None):
+        # seeds (optional): equal-length per-proof subset seeds — each
+        # gossip verify samples with its own peer seed
+        # (validation.go:206-209).
+        vc, _seed = self._vc(opts)
+        cps, cms, n = (marshalled if marshalled is not None
+                       else self.marshal_batch(proofs, metas))
         statuses = (c_int * n)()
         invs = (c_uint32 * n)()
         if seeds is not None:
