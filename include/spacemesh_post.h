@@ -109,7 +109,10 @@ typedef struct {
 typedef struct PostInitSession PostInitSession;
 
 /* Create a session. Scans data_dir for existing postdata_*.bin and resumes
- * after the last complete label (StartSession resume semantics,
+ * after the last complete label; the persisted metadata Nonce/NonceValue
+ * (re-written whenever the running VRF minimum improves) seeds the
+ * session's minimum, so sharded sessions keep their shard-local minimum
+ * across kill+resume (StartSession resume semantics,
  * activation/post.go:267-271). */
 int post_init_new(const PostInitConfig *cfg, PostInitSession **out);
 
