@@ -215,7 +215,7 @@ extern "C" {
 const char *post_last_error(void) { return g_last_error.c_str(); }
 
 const char *post_engine_version(void) {
-  return "spacemesh-post-hip 0.1 (gfx950)";
+  return "spacemesh-post-hip 0.2 (gfx950)";
 }
 
 /* ------------------------- providers ------------------------- */

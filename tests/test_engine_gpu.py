@@ -505,6 +505,26 @@ def test_concurrent_mixed_ops(roundtrip, tmp_path):
     assert all(ok for _, ok in results), results
 
 
+def test_scan_hit_overflow_is_loud():
+    """A pathological K1/num_labels configuration that makes every label
+    pass every nonce must fail loudly (scan hit-buffer overflow), never
+    return a silently-truncated proof."""
+    total = 1 << 21
+    cfg, mgr = make_mgr(1, total, 32)
+    mgr.prepare_initializer()
+    mgr.start_session()
+    labels = mgr.copy_labels(0, total)
+    mgr.reset()
+    # k1 == num_labels -> proving difficulty ~2^64: all 288*2 nonce values
+    # of every label hit = 1.2e9 hits >> the 4M hit buffer
+    bad = gsm_amd.PostConfig(min_num_units=1, labels_per_unit=total,
+                             k1=total, k2=37, pow_difficulty=POW_DIFF)
+    with pytest.raises(gsm_amd.EngineError) as ei:
+        gsm_amd.api.prove_buffer(labels, total, NODE, ATX, CHALLENGE, bad,
+                                 gsm_amd.ProveOpts(nonces=288))
+    assert "overflow" in str(ei.value)
+
+
 def test_selfcheck_period_preserves_parity():
     """POST_SELFCHECK_PERIOD thins the host reference-label check (multi-
     rank contention knob) but must not change any output byte."""
