@@ -31,16 +31,21 @@ def test_merge_nonces_min_label_then_index():
     assert sharding.merge_nonces([a]) == a
 
 
-def _gloo_worker(rank, world, tmpdir):
+def _gloo_worker(rank, world, port, nonce_holders):
     import torch.distributed as dist
     os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = "29512"
+    os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
-    # each rank contributes its shard-local candidate
-    local = (rank * 100 + 1, bytes([rank + 1]) + bytes(31))
+    # each rank contributes its shard-local candidate; some ranks may have
+    # found none (nonce optional per shard — bench.py passes None then)
+    if rank in nonce_holders:
+        local = (rank * 100 + 1, bytes([rank + 1]) + bytes(31))
+    else:
+        local = None
     got = sharding.allreduce_nonce(local)
-    # rank 0's label (0x01...) is the global minimum
-    assert got == (1, bytes([1]) + bytes(31)), got
+    # the lowest-numbered holder's label is the global minimum
+    lo = min(nonce_holders)
+    assert got == (lo * 100 + 1, bytes([lo + 1]) + bytes(31)), got
     # the bench's whole-job MAX-over-ranks reduction (bench.py timed region)
     import torch
     t = torch.tensor([float(rank + 1)], dtype=torch.float64)
@@ -50,13 +55,25 @@ def _gloo_worker(rank, world, tmpdir):
     dist.destroy_process_group()
 
 
-def test_allreduce_nonce_gloo_world2(tmp_path):
+def _run_world(world, port, nonce_holders):
     import torch.multiprocessing as mp
     ctx = mp.get_context("spawn")
-    procs = [ctx.Process(target=_gloo_worker, args=(r, 2, str(tmp_path)))
-             for r in range(2)]
+    procs = [ctx.Process(target=_gloo_worker,
+                         args=(r, world, port, nonce_holders))
+             for r in range(world)]
     for p in procs:
         p.start()
     for p in procs:
-        p.join(120)
+        p.join(180)
         assert p.exitcode == 0
+
+
+def test_allreduce_nonce_gloo_world2():
+    _run_world(2, 29512, {0, 1})
+
+
+def test_allreduce_nonce_gloo_world8_sparse():
+    """The driver's SCALE shape is 8 ranks; some shards may finish without
+    a below-threshold candidate, so the min-reduce must also tolerate
+    rank-local None contributions."""
+    _run_world(8, 29513, {3, 5})
